@@ -1,0 +1,6 @@
+"""Client layer: autograd-integrated remote expert calls."""
+
+from .expert import RemoteExpert, rpc_call
+from .moe import RemoteMixtureOfExperts
+
+__all__ = ["RemoteExpert", "RemoteMixtureOfExperts", "rpc_call"]

@@ -1,0 +1,276 @@
+"""RemoteMixtureOfExperts — DMoE client layer (SURVEY.md §2
+"RemoteMixtureOfExperts", §3.1 hot path).
+
+Semantics preserved from the reference:
+1. gating: one Linear produces per-grid-dimension logits; a full uid
+   "prefix.i.j" scores as the sum of its per-dimension logits;
+2. per-token top-k expert selection, filtered through the DHT for experts
+   that are ACTIVE right now (liveness lease);
+3. parallel dispatch with a quorum: proceed when >= k_min experts responded
+   within the timeout, DROP the rest;
+4. combine: softmax over the logits of the experts that actually responded
+   (renormalization over survivors), weighted sum of their outputs;
+5. backward mirrors 3-4: parallel "bwd" RPCs, dead experts contribute zero.
+
+MI355X: this socket-RPC path is the cross-host / parity mode. The intra-node
+hot path implements the same math with RCCL all-to-all + CDNA4 kernels in
+parallel/expert_shard.py. The per-token selection logic here batches all DHT
+probes per forward (one get_experts over the candidate union, cached for
+`active_cache_ttl`) instead of per-sample DHT walks — same observable
+semantics, control-plane cost O(unique candidates) not O(batch).
+"""
+from __future__ import annotations
+
+import time
+from concurrent.futures import ThreadPoolExecutor, wait as futures_wait
+from typing import Dict, List, Optional, Sequence, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from ..dht import DHT, UID_DELIMITER
+from ..utils import Endpoint, get_logger
+from .expert import rpc_call
+
+logger = get_logger(__name__)
+
+
+class RemoteMixtureOfExperts(nn.Module):
+    def __init__(
+        self,
+        *,
+        in_features: int,
+        grid_size: Sequence[int],
+        dht: DHT,
+        uid_prefix: str,
+        k_best: int,
+        k_min: int = 1,
+        forward_timeout: float = 10.0,
+        backward_timeout: float = 10.0,
+        backward_k_min: int = 1,
+        active_cache_ttl: float = 10.0,
+        candidate_factor: int = 4,
+        aux_loss_coef: float = 0.0,
+    ):
+        super().__init__()
+        self.grid_size = tuple(grid_size)
+        self.dht = dht
+        self.uid_prefix = uid_prefix
+        self.k_best, self.k_min, self.backward_k_min = k_best, k_min, backward_k_min
+        self.forward_timeout, self.backward_timeout = forward_timeout, backward_timeout
+        self.active_cache_ttl = active_cache_ttl
+        self.candidate_factor = candidate_factor
+        self.aux_loss_coef = aux_loss_coef
+        self.proj = nn.Linear(in_features, sum(self.grid_size))
+        self._active_cache: Dict[str, Optional[Endpoint]] = {}
+        self._active_cache_time = -float("inf")
+        self.aux_loss: Optional[torch.Tensor] = None
+
+    # ------------------------------------------------------------------
+    def _candidate_uids(self, grid_logits: List[torch.Tensor]) -> Tuple[List[List[str]], torch.Tensor]:
+        """Local beam over grid dims: per sample, the top
+        (candidate_factor * k_best) full uids by summed per-dim logits.
+        Returns (uid rows, scores [N, C])."""
+        beam_width = min(self.candidate_factor * self.k_best,
+                         int(torch.tensor(self.grid_size).prod()))
+        N = grid_logits[0].shape[0]
+        # beam state: scores [N, B], index tuples per sample
+        scores = grid_logits[0]  # [N, grid_size[0]]
+        topv, topi = scores.topk(min(beam_width, scores.shape[1]), dim=-1)
+        beams: List[List[Tuple[int, ...]]] = [[(int(i),) for i in topi[n]] for n in range(N)]
+        beam_scores = topv
+        for dim in range(1, len(self.grid_size)):
+            dim_logits = grid_logits[dim]  # [N, grid_size[dim]]
+            dim_top_v, dim_top_i = dim_logits.topk(min(beam_width, dim_logits.shape[1]), dim=-1)
+            combined = beam_scores.unsqueeze(-1) + dim_top_v.unsqueeze(1)  # [N, B, B']
+            flat = combined.flatten(1)
+            topv, topi = flat.topk(min(beam_width, flat.shape[1]), dim=-1)
+            new_beams = []
+            for n in range(N):
+                row = []
+                for flat_idx in topi[n].tolist():
+                    b, bp = divmod(flat_idx, dim_top_i.shape[1])
+                    row.append(beams[n][b] + (int(dim_top_i[n, bp]),))
+                new_beams.append(row)
+            beams, beam_scores = new_beams, topv
+        uid_rows = [
+            [self.uid_prefix + UID_DELIMITER + UID_DELIMITER.join(map(str, idxs)) for idxs in row]
+            for row in beams
+        ]
+        return uid_rows, beam_scores
+
+    def _resolve_active(self, uids: Sequence[str]) -> Dict[str, Optional[Endpoint]]:
+        now = time.monotonic()
+        if now - self._active_cache_time > self.active_cache_ttl:
+            self._active_cache.clear()
+            self._active_cache_time = now
+        missing = [u for u in uids if u not in self._active_cache]
+        if missing:
+            endpoints = self.dht.get_experts(missing)
+            for uid, ep in zip(missing, endpoints):
+                self._active_cache[uid] = ep
+        return {u: self._active_cache.get(u) for u in uids}
+
+    # ------------------------------------------------------------------
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        lead_shape, d = x.shape[:-1], x.shape[-1]
+        flat = x.reshape(-1, d)
+        N = flat.shape[0]
+        all_logits = self.proj(flat)
+        grid_logits = list(all_logits.split(list(self.grid_size), dim=-1))
+
+        uid_rows, cand_scores = self._candidate_uids(grid_logits)
+        unique_uids = sorted({u for row in uid_rows for u in row})
+        active = self._resolve_active(unique_uids)
+
+        # per sample: top-k_best ACTIVE candidates
+        sel_uids: List[List[str]] = []
+        sel_cols: List[List[int]] = []
+        for n in range(N):
+            chosen, cols = [], []
+            for c, uid in enumerate(uid_rows[n]):
+                if active.get(uid) is not None:
+                    chosen.append(uid)
+                    cols.append(c)
+                    if len(chosen) >= self.k_best:
+                        break
+            if not chosen:
+                raise RuntimeError(
+                    f"RemoteMixtureOfExperts: no active experts among "
+                    f"{len(uid_rows[n])} candidates (prefix={self.uid_prefix!r})"
+                )
+            sel_uids.append(chosen)
+            sel_cols.append(cols)
+        k_max = max(len(row) for row in sel_uids)
+        sel_logits = torch.full((N, k_max), float("-inf"), dtype=cand_scores.dtype,
+                                device=cand_scores.device)
+        for n in range(N):
+            sel_logits[n, : len(sel_cols[n])] = cand_scores[n, sel_cols[n]]
+
+        endpoint_of = {u: active[u] for row in sel_uids for u in row}
+        y = _RemoteMoECallBatched.apply(
+            flat, sel_logits, sel_uids, endpoint_of, self.k_min,
+            self.forward_timeout, self.backward_timeout,
+        )
+
+        if self.aux_loss_coef > 0:
+            self.aux_loss = self.aux_loss_coef * _importance_aux_loss(grid_logits)
+        else:
+            self.aux_loss = None
+        return y.reshape(*lead_shape, -1)
+
+
+def _importance_aux_loss(grid_logits: List[torch.Tensor]) -> torch.Tensor:
+    """Shazeer-style importance penalty: squared coefficient of variation of
+    per-expert-index softmax mass, summed over grid dimensions."""
+    total = 0.0
+    for logits in grid_logits:
+        probs = F.softmax(logits, dim=-1)
+        importance = probs.sum(0)
+        cv_sq = importance.var(unbiased=False) / (importance.mean() ** 2 + 1e-10)
+        total = total + cv_sq
+    return total
+
+
+class _RemoteMoECallBatched(torch.autograd.Function):
+    """Batched dispatch/combine with quorum fault tolerance and parallel
+    backward RPCs (manual grads: softmax-over-survivors combine)."""
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, sel_logits: torch.Tensor, sel_uids: List[List[str]],
+                endpoint_of: Dict[str, Endpoint], k_min: int,
+                forward_timeout: float, backward_timeout: float):
+        N, d = x.shape
+        k_max = sel_logits.shape[1]
+        x_cpu = x.detach().cpu()
+
+        # group (token, slot) pairs per expert uid
+        expert_tokens: Dict[str, List[Tuple[int, int]]] = {}
+        for n, row in enumerate(sel_uids):
+            for slot, uid in enumerate(row):
+                expert_tokens.setdefault(uid, []).append((n, slot))
+
+        outputs: Dict[str, torch.Tensor] = {}
+        with ThreadPoolExecutor(max_workers=min(32, len(expert_tokens))) as pool:
+            futures = {}
+            for uid, pairs in expert_tokens.items():
+                idx = [n for n, _ in pairs]
+                futures[uid] = pool.submit(
+                    rpc_call, endpoint_of[uid], "fwd", [uid, [x_cpu[idx]]], forward_timeout
+                )
+            done, not_done = futures_wait(futures.values(), timeout=forward_timeout)
+            for uid, fut in futures.items():
+                if fut in done and fut.exception() is None:
+                    outputs[uid] = fut.result()[0]
+                else:
+                    fut.cancel()
+                    logger.warning(f"forward RPC to expert {uid} failed/timed out; dropping it")
+
+        if not outputs:
+            raise RuntimeError("RemoteMixtureOfExperts: every selected expert failed")
+        d_out = next(iter(outputs.values())).shape[-1]
+        o = torch.zeros(N, k_max, d_out, dtype=x.dtype)
+        responded = torch.zeros(N, k_max, dtype=torch.bool)
+        for uid, out in outputs.items():
+            for local_i, (n, slot) in enumerate(expert_tokens[uid]):
+                o[n, slot] = out[local_i].to(o.dtype)
+                responded[n, slot] = True
+
+        quorum = responded.sum(1)
+        if (quorum < k_min).any():
+            bad = int((quorum < k_min).sum())
+            raise RuntimeError(
+                f"RemoteMixtureOfExperts: {bad}/{N} tokens got fewer than k_min={k_min} expert replies"
+            )
+
+        masked_logits = sel_logits.detach().cpu().float().masked_fill(~responded, float("-inf"))
+        w = F.softmax(masked_logits, dim=-1)  # softmax over survivors
+        w = torch.nan_to_num(w, nan=0.0)
+        y = torch.einsum("nk,nkd->nd", w.to(o.dtype), o)
+
+        ctx.save_for_backward(x_cpu, o, w, responded)
+        ctx.expert_tokens = expert_tokens
+        ctx.responded_uids = set(outputs.keys())
+        ctx.endpoint_of = endpoint_of
+        ctx.backward_timeout = backward_timeout
+        ctx.device = x.device
+        return y.to(x.device)
+
+    @staticmethod
+    def backward(ctx, grad_y: torch.Tensor):
+        x_cpu, o, w, responded = ctx.saved_tensors
+        g = grad_y.detach().cpu().to(o.dtype)
+        N, k_max, d_out = o.shape
+
+        # grads wrt gating logits through softmax-over-survivors
+        dot = torch.einsum("nkd,nd->nk", o.float(), g.float())  # <o_nk, g_n>
+        weighted = (w * dot).sum(-1, keepdim=True)
+        dlogits = (w * (dot - weighted)).masked_fill(~responded, 0.0)
+
+        # grads wrt x: parallel bwd RPCs, one per expert that responded in fwd
+        dx = torch.zeros_like(x_cpu, dtype=torch.float32)
+        live_experts = [uid for uid in ctx.expert_tokens if uid in ctx.responded_uids]
+        with ThreadPoolExecutor(max_workers=min(32, max(1, len(live_experts)))) as pool:
+            futures = {}
+            for uid in live_experts:
+                pairs = ctx.expert_tokens[uid]
+                idx = [n for n, _ in pairs]
+                slots = [s for _, s in pairs]
+                grad_out = w[idx, slots].unsqueeze(1).to(g.dtype) * g[idx]
+                futures[uid] = (
+                    pool.submit(rpc_call, ctx.endpoint_of[uid], "bwd",
+                                [uid, [x_cpu[idx], grad_out]], ctx.backward_timeout),
+                    idx,
+                )
+            done, _ = futures_wait([f for f, _ in futures.values()], timeout=ctx.backward_timeout)
+            for uid, (fut, idx) in futures.items():
+                if fut in done and fut.exception() is None:
+                    grad_x_part = fut.result()[0].float()
+                    dx[idx] += grad_x_part
+                else:
+                    fut.cancel()
+                    logger.warning(f"backward RPC to expert {uid} failed/timed out; zero grads")
+
+        return (dx.to(ctx.device, dtype=grad_y.dtype),
+                dlogits.to(ctx.device), None, None, None, None, None)

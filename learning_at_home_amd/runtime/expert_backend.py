@@ -1,0 +1,120 @@
+"""ExpertBackend — one trainable expert behind stateless-RPC-shaped
+forward/backward (SURVEY.md §2 "ExpertBackend", §3.1 call stack).
+
+Key reference semantics preserved exactly:
+- `forward` keeps NO activations between calls;
+- `backward` RE-RUNS forward under `enable_grad` on the inputs the client
+  sends back (recompute-in-backward), runs autograd, returns grad_inputs;
+- after every backward batch the expert's own optimizer steps immediately
+  (asynchronous / delayed-gradient SGD — no global barrier anywhere).
+
+MI355X path: when the wrapped module is a `models.FFNExpertHIP` (or any
+module exposing `.fused_forward/.fused_backward`), compute runs through the
+hand-written CDNA4 grouped-FFN kernels; otherwise plain PyTorch ops (CPU
+tests, config #1 in BASELINE.json).
+"""
+from __future__ import annotations
+
+from typing import Any, Callable, Dict, Optional, Sequence, Tuple
+
+import torch
+import torch.nn as nn
+
+from ..utils import BatchTensorDescriptor, get_logger
+from .task_pool import TaskPool
+
+logger = get_logger(__name__)
+
+
+class ExpertBackend:
+    def __init__(
+        self,
+        name: str,
+        expert: nn.Module,
+        optimizer: torch.optim.Optimizer,
+        *,
+        args_schema: Sequence[BatchTensorDescriptor],
+        outputs_schema: Optional[Sequence[BatchTensorDescriptor]] = None,
+        max_batch_size: int = 16384,
+        grad_clip: Optional[float] = None,
+    ):
+        self.name = name
+        self.expert = expert
+        self.optimizer = optimizer
+        self.args_schema = tuple(args_schema)
+        self.grad_clip = grad_clip
+        self.update_count = 0
+
+        if outputs_schema is None:
+            with torch.no_grad():
+                dummy = [schema.make_batch(1) for schema in self.args_schema]
+                device = next(expert.parameters()).device if any(True for _ in expert.parameters()) else torch.device("cpu")
+                dummy = [d.to(device) for d in dummy]
+                out = expert(*dummy)
+                out = out if isinstance(out, tuple) else (out,)
+                outputs_schema = tuple(BatchTensorDescriptor.from_tensor(o) for o in out)
+        self.outputs_schema = tuple(outputs_schema)
+
+        self.forward_pool = TaskPool(self.forward, f"{name}_forward", max_batch_size)
+        self.backward_pool = TaskPool(self.backward, f"{name}_backward", max_batch_size)
+
+    # ------------------------------------------------------------------
+    def forward(self, *inputs: torch.Tensor) -> Tuple[torch.Tensor, ...]:
+        """Inference-mode forward. No state is kept (backward recomputes)."""
+        with torch.no_grad():
+            outputs = self.expert(*inputs)
+        return outputs if isinstance(outputs, tuple) else (outputs,)
+
+    def backward(self, *inputs_and_grads: torch.Tensor) -> Tuple[torch.Tensor, ...]:
+        """Recompute forward under enable_grad, backprop the provided output
+        grads, step this expert's optimizer, and return input grads."""
+        num_inputs = len(self.args_schema)
+        inputs, grad_outputs = inputs_and_grads[:num_inputs], inputs_and_grads[num_inputs:]
+        with torch.enable_grad():
+            detached = [x.detach().requires_grad_(x.dtype.is_floating_point) for x in inputs]
+            outputs = self.expert(*detached)
+            outputs = outputs if isinstance(outputs, tuple) else (outputs,)
+            assert len(outputs) == len(grad_outputs), (
+                f"{self.name}: got {len(grad_outputs)} grad tensors for {len(outputs)} outputs"
+            )
+            torch.autograd.backward(
+                outputs, grad_tensors=[g.to(o.dtype) for g, o in zip(grad_outputs, outputs)],
+                retain_graph=False, create_graph=False,
+            )
+        grad_inputs = tuple(
+            x.grad if x.grad is not None else torch.zeros_like(x) for x in detached
+        )
+        self.apply_gradients()
+        return grad_inputs
+
+    def apply_gradients(self) -> None:
+        """One asynchronous optimizer step on this expert's own parameters."""
+        if self.grad_clip is not None:
+            torch.nn.utils.clip_grad_norm_(self.expert.parameters(), self.grad_clip)
+        self.optimizer.step()
+        self.optimizer.zero_grad(set_to_none=True)
+        self.update_count += 1
+
+    # ------------------------------------------------------------------
+    def get_info(self) -> Dict[str, Any]:
+        return {
+            "name": self.name,
+            "args_schema": [s.to_dict() for s in self.args_schema],
+            "outputs_schema": [s.to_dict() for s in self.outputs_schema],
+        }
+
+    def get_pools(self) -> Tuple[TaskPool, TaskPool]:
+        return self.forward_pool, self.backward_pool
+
+    # checkpointing (SURVEY.md §5 "Checkpoint / resume": byte-compatible torch.save maps)
+    def state_dict(self) -> Dict[str, Any]:
+        return {
+            "model": self.expert.state_dict(),
+            "optimizer": self.optimizer.state_dict(),
+            "update_count": self.update_count,
+        }
+
+    def load_state_dict(self, state: Dict[str, Any]) -> None:
+        self.expert.load_state_dict(state["model"])
+        self.optimizer.load_state_dict(state["optimizer"])
+        self.update_count = int(state.get("update_count", 0))

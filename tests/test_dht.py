@@ -1,0 +1,133 @@
+"""DHT tests (SURVEY.md §4 "DHT tests"): real Kademlia peers on localhost —
+store/get across peers, expiration, expert declaration, first_k_active,
+node death tolerance. No mocks."""
+import time
+
+import pytest
+
+from learning_at_home_amd.dht import DHT, DHTID, uid_prefixes
+from learning_at_home_amd.dht.routing import RoutingTable
+from learning_at_home_amd.utils import get_dht_time
+
+
+@pytest.fixture
+def dht_pair():
+    first = DHT(start=True)
+    second = DHT(initial_peers=[first.endpoint], start=True)
+    yield first, second
+    second.shutdown()
+    first.shutdown()
+
+
+class TestRouting:
+    def test_dhtid_deterministic(self):
+        assert DHTID.generate("key") == DHTID.generate("key")
+        assert DHTID.generate("key") != DHTID.generate("yek")
+
+    def test_dhtid_bytes_roundtrip(self):
+        i = DHTID.generate("x")
+        assert DHTID.from_bytes(i.to_bytes()) == i
+
+    def test_routing_table_nearest(self):
+        me = DHTID.generate("me")
+        table = RoutingTable(me, bucket_size=4)
+        ids = [DHTID.generate(f"node{i}") for i in range(64)]
+        for nid in ids:
+            table.add_or_update_node(int(nid), ("127.0.0.1", 1000 + int(nid) % 1000))
+        query = DHTID.generate("query")
+        nearest = table.get_nearest_neighbors(int(query), k=8)
+        assert len(nearest) == 8
+        dists = [int(query) ^ nid for nid, _ in nearest]
+        assert dists == sorted(dists)
+        # nearest-of-returned must be the global nearest among stored nodes
+        stored = [nid for b in table.buckets for nid in b.nodes]
+        assert min(dists) == min(int(query) ^ n for n in stored)
+
+
+class TestDHTProcess:
+    def test_single_node_store_get(self):
+        dht = DHT(start=True)
+        try:
+            assert dht.store("solo", [1, 2], get_dht_time() + 30)
+            value, exp = dht.get("solo")
+            assert value == [1, 2]
+        finally:
+            dht.shutdown()
+
+    def test_two_node_store_get(self, dht_pair):
+        first, second = dht_pair
+        assert first.store("shared_key", "hello", get_dht_time() + 30)
+        value, exp = second.get("shared_key")
+        assert value == "hello"
+
+    def test_expiration_across_nodes(self, dht_pair):
+        first, second = dht_pair
+        first.store("ephemeral", "x", get_dht_time() + 0.5)
+        assert second.get("ephemeral")[0] == "x"
+        time.sleep(0.8)
+        fresh = DHT(initial_peers=[first.endpoint], start=True)  # no cache
+        try:
+            assert fresh.get("ephemeral")[0] is None
+        finally:
+            fresh.shutdown()
+
+    def test_many_nodes(self):
+        """A dozen peers bootstrap off one seed; any peer sees any store."""
+        seed = DHT(start=True)
+        peers = [DHT(initial_peers=[seed.endpoint], start=True) for _ in range(11)]
+        try:
+            for i, peer in enumerate(peers[:4]):
+                assert peer.store(f"key{i}", i * 100, get_dht_time() + 60)
+            for i in range(4):
+                value, _ = peers[-1 - i].get(f"key{i}")
+                assert value == i * 100, f"key{i} not found from peer {-1 - i}"
+        finally:
+            for p in peers:
+                p.shutdown()
+            seed.shutdown()
+
+    def test_node_death_does_not_break_lookup(self):
+        seed = DHT(start=True)
+        peers = [DHT(initial_peers=[seed.endpoint], start=True) for _ in range(4)]
+        try:
+            peers[0].store("survivor_key", "v", get_dht_time() + 60)
+            peers[1].shutdown()  # kill one peer
+            value, _ = peers[3].get("survivor_key")
+            assert value == "v"
+        finally:
+            for p in (peers[0], peers[2], peers[3]):
+                p.shutdown()
+            seed.shutdown()
+
+
+class TestExpertDiscovery:
+    def test_uid_prefixes(self):
+        assert uid_prefixes("ffn.3.7") == ["ffn", "ffn.3"]
+        assert uid_prefixes("ffn") == []
+
+    def test_declare_and_get_experts(self, dht_pair):
+        first, second = dht_pair
+        uids = ["ffn.0", "ffn.1", "ffn.5"]
+        assert first.declare_experts(uids, "127.0.0.1", 9000, expiration=60)
+        endpoints = second.get_experts(["ffn.1", "ffn.99", "ffn.5"])
+        assert endpoints[0] == ("127.0.0.1", 9000)
+        assert endpoints[1] is None
+        assert endpoints[2] == ("127.0.0.1", 9000)
+
+    def test_first_k_active(self, dht_pair):
+        first, second = dht_pair
+        first.declare_experts(["ffn.0.1", "ffn.2.3"], "127.0.0.1", 9001, expiration=60)
+        active = second.first_k_active(["ffn.7", "ffn.0", "ffn.2", "ffn.9"], k=2)
+        assert list(active.keys()) == ["ffn.0", "ffn.2"]
+        assert active["ffn.0"] == ("127.0.0.1", 9001)
+
+    def test_lease_expires(self, dht_pair):
+        first, second = dht_pair
+        first.declare_experts(["gone.0"], "127.0.0.1", 9002, expiration=0.5)
+        assert second.get_experts(["gone.0"])[0] is not None
+        time.sleep(0.8)
+        fresh = DHT(initial_peers=[first.endpoint], start=True)
+        try:
+            assert fresh.get_experts(["gone.0"])[0] is None
+        finally:
+            fresh.shutdown()

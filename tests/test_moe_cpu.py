@@ -1,0 +1,239 @@
+"""Integration tests (SURVEY.md §4 "Integration, single machine"): real
+Server.create(...) with CPU experts + real DHT peers on 127.0.0.1, no mocks.
+This is BASELINE.json config #1: 4-expert FFN DMoE, d_model=128."""
+import time
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+from learning_at_home_amd import DHT, RemoteExpert, RemoteMixtureOfExperts, Server
+from learning_at_home_amd.models import FFNExpert
+
+D_MODEL = 128
+
+
+@pytest.fixture(scope="module")
+def moe_grid():
+    """One DHT seed + 2 servers hosting 2 experts each on a 'ffn.<i>' grid."""
+    dht = DHT(start=True)
+    servers = []
+    for offset in (0, 2):
+        server = Server.create(
+            num_experts=2,
+            expert_cls="ffn",
+            hidden_dim=D_MODEL,
+            expert_prefix="ffn",
+            expert_offset=offset,
+            initial_peers=[dht.endpoint],
+            announce_period=1.0,
+            start=True,
+        )
+        servers.append(server)
+    client_dht = DHT(initial_peers=[dht.endpoint], start=True)
+    yield dht, servers, client_dht
+    client_dht.shutdown()
+    for s in servers:
+        s.shutdown()
+        if s.dht is not None:
+            s.dht.shutdown()
+    dht.shutdown()
+
+
+class TestRemoteExpert:
+    def test_forward_matches_local(self, moe_grid):
+        _, servers, client_dht = moe_grid
+        server = servers[0]
+        uid = next(iter(server.expert_backends))
+        local_module = server.expert_backends[uid].expert
+        (endpoint,) = client_dht.get_experts([uid])
+        assert endpoint is not None, "expert not discoverable via DHT"
+        remote = RemoteExpert(uid, *endpoint)
+
+        x = torch.randn(5, D_MODEL)
+        with torch.no_grad():
+            expected = local_module(x)
+        out = remote(x)
+        assert torch.allclose(out, expected, atol=1e-5), "remote forward != local module"
+
+    def test_gradients_flow(self, moe_grid):
+        _, servers, client_dht = moe_grid
+        uid = next(iter(servers[0].expert_backends))
+        (endpoint,) = client_dht.get_experts([uid])
+        remote = RemoteExpert(uid, *endpoint)
+        x = torch.randn(3, D_MODEL, requires_grad=True)
+        out = remote(x)
+        out.sum().backward()
+        assert x.grad is not None and x.grad.abs().sum() > 0
+
+    def test_grad_matches_local(self, moe_grid):
+        _, servers, client_dht = moe_grid
+        uid = next(iter(servers[0].expert_backends))
+        backend = servers[0].expert_backends[uid]
+        # snapshot weights BEFORE backward (backward applies an async SGD step)
+        local_copy = FFNExpert(D_MODEL)
+        local_copy.load_state_dict(backend.expert.state_dict())
+        (endpoint,) = client_dht.get_experts([uid])
+        remote = RemoteExpert(uid, *endpoint)
+
+        x = torch.randn(4, D_MODEL, requires_grad=True)
+        remote(x).pow(2).sum().backward()
+        x_ref = x.detach().clone().requires_grad_(True)
+        local_copy(x_ref).pow(2).sum().backward()
+        assert torch.allclose(x.grad, x_ref.grad, atol=1e-4)
+
+    def test_backward_applies_async_sgd_step(self, moe_grid):
+        _, servers, client_dht = moe_grid
+        uid = list(servers[1].expert_backends)[0]
+        backend = servers[1].expert_backends[uid]
+        before = backend.update_count
+        (endpoint,) = client_dht.get_experts([uid])
+        remote = RemoteExpert(uid, *endpoint)
+        x = torch.randn(2, D_MODEL, requires_grad=True)
+        remote(x).sum().backward()
+        deadline = time.monotonic() + 5
+        while backend.update_count == before and time.monotonic() < deadline:
+            time.sleep(0.05)
+        assert backend.update_count == before + 1, "server did not apply async SGD step"
+
+    def test_info_schema(self, moe_grid):
+        _, servers, client_dht = moe_grid
+        uid = next(iter(servers[0].expert_backends))
+        (endpoint,) = client_dht.get_experts([uid])
+        remote = RemoteExpert(uid, *endpoint)
+        assert remote.info["name"] == uid
+        assert remote.args_schema[0].shape[1:] == (D_MODEL,)
+
+
+class TestRemoteMixtureOfExperts:
+    def _make_moe(self, client_dht, **kwargs):
+        defaults = dict(
+            in_features=D_MODEL, grid_size=(4,), dht=client_dht, uid_prefix="ffn",
+            k_best=2, k_min=1, forward_timeout=10.0, backward_timeout=10.0,
+            active_cache_ttl=1.0,
+        )
+        defaults.update(kwargs)
+        return RemoteMixtureOfExperts(**defaults)
+
+    def test_forward_backward(self, moe_grid):
+        _, _, client_dht = moe_grid
+        moe = self._make_moe(client_dht)
+        x = torch.randn(6, D_MODEL, requires_grad=True)
+        y = moe(x)
+        assert y.shape == (6, D_MODEL)
+        y.sum().backward()
+        assert x.grad is not None and x.grad.abs().sum() > 0
+        assert moe.proj.weight.grad is not None, "gating projection got no grads"
+
+    def test_forward_matches_manual_combine(self, moe_grid):
+        """MoE output == softmax-weighted sum of the chosen experts' outputs."""
+        _, servers, client_dht = moe_grid
+        moe = self._make_moe(client_dht, k_best=4)
+        local = {}
+        for server in servers:
+            for uid, backend in server.expert_backends.items():
+                m = FFNExpert(D_MODEL)
+                m.load_state_dict(backend.expert.state_dict())
+                local[uid] = m
+        x = torch.randn(3, D_MODEL)
+        with torch.no_grad():
+            y = moe(x)
+            logits = moe.proj(x)
+            w = F.softmax(logits, dim=-1)
+            expected = torch.zeros_like(x)
+            for e, uid in enumerate(sorted(local, key=lambda u: int(u.split(".")[1]))):
+                expected += w[:, e : e + 1] * local[uid](x)
+        assert torch.allclose(y, expected, atol=1e-4), (y - expected).abs().max()
+
+    def test_3d_input(self, moe_grid):
+        _, _, client_dht = moe_grid
+        moe = self._make_moe(client_dht)
+        x = torch.randn(2, 5, D_MODEL)
+        assert moe(x).shape == (2, 5, D_MODEL)
+
+    def test_aux_loss(self, moe_grid):
+        _, _, client_dht = moe_grid
+        moe = self._make_moe(client_dht, aux_loss_coef=0.01)
+        x = torch.randn(4, D_MODEL, requires_grad=True)
+        y = moe(x)
+        assert moe.aux_loss is not None and moe.aux_loss.requires_grad
+        (y.sum() + moe.aux_loss).backward()
+
+
+class TestFaultTolerance:
+    def test_moe_survives_dead_server(self):
+        """Kill one of two servers: MoE keeps working on surviving experts
+        (expert dropout + renormalize — SURVEY.md §5)."""
+        dht = DHT(start=True)
+        s1 = Server.create(2, "ffn", D_MODEL, expert_prefix="tol", expert_offset=0,
+                           initial_peers=[dht.endpoint], announce_period=0.5, start=True)
+        s2 = Server.create(2, "ffn", D_MODEL, expert_prefix="tol", expert_offset=2,
+                           initial_peers=[dht.endpoint], announce_period=0.5, start=True)
+        client_dht = DHT(initial_peers=[dht.endpoint], start=True)
+        try:
+            moe = RemoteMixtureOfExperts(
+                in_features=D_MODEL, grid_size=(4,), dht=client_dht, uid_prefix="tol",
+                k_best=4, k_min=1, forward_timeout=3.0, backward_timeout=3.0,
+                active_cache_ttl=0.2,
+            )
+            x = torch.randn(3, D_MODEL, requires_grad=True)
+            y0 = moe(x)
+            assert y0.shape == (3, D_MODEL)
+            # kill server 2 (its experts stop responding; lease expires in ~1s)
+            s2.shutdown()
+            if s2.dht is not None:
+                s2.dht.shutdown()
+            time.sleep(1.5)
+            # the first lookup after a death may transiently see nothing
+            # (timeout discovery window); the system must RECOVER within a
+            # few seconds using only s1's experts
+            deadline = time.monotonic() + 15
+            y1 = None
+            while time.monotonic() < deadline:
+                try:
+                    y1 = moe(x)
+                    break
+                except RuntimeError:
+                    time.sleep(0.5)
+            assert y1 is not None, "MoE did not recover after server death"
+            assert y1.shape == (3, D_MODEL)
+            y1.sum().backward()
+            assert x.grad is not None
+        finally:
+            client_dht.shutdown()
+            s1.shutdown()
+            if s1.dht is not None:
+                s1.dht.shutdown()
+            dht.shutdown()
+
+    def test_restarted_server_rediscovered(self):
+        """A server that dies and a NEW server that declares the same uid is
+        picked up after the active-cache TTL (re-announce contract)."""
+        dht = DHT(start=True)
+        client_dht = DHT(initial_peers=[dht.endpoint], start=True)
+        s1 = Server.create(1, "ffn", D_MODEL, expert_prefix="re", expert_offset=0,
+                           initial_peers=[dht.endpoint], announce_period=0.5, start=True)
+        try:
+            (ep1,) = client_dht.get_experts(["re.0"])
+            assert ep1 is not None
+            s1.shutdown()
+            s2 = Server.create(1, "ffn", D_MODEL, expert_prefix="re", expert_offset=0,
+                               initial_peers=[dht.endpoint], announce_period=0.5, start=True)
+            try:
+                deadline = time.monotonic() + 10
+                ep2 = None
+                while time.monotonic() < deadline:
+                    (ep2,) = client_dht.get_experts(["re.0"])
+                    if ep2 is not None and tuple(ep2) != tuple(ep1):
+                        break
+                    time.sleep(0.2)
+                assert ep2 is not None and tuple(ep2) != tuple(ep1), "restarted server not rediscovered"
+            finally:
+                s2.shutdown()
+                if s2.dht is not None:
+                    s2.dht.shutdown()
+        finally:
+            if s1.dht is not None:
+                s1.dht.shutdown()
+            client_dht.shutdown()
+            dht.shutdown()

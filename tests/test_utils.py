@@ -1,0 +1,142 @@
+"""Unit tests for L0 utilities (serialization round-trip,
+nested pack/unpack, MPFuture pair semantics, TimedStorage expiration)."""
+import threading
+import time
+from concurrent.futures import TimeoutError as FutureTimeoutError
+
+import pytest
+import torch
+
+from learning_at_home_amd.utils import (
+    MPFuture,
+    MSGPackSerializer,
+    PytorchSerializer,
+    TimedStorage,
+    get_dht_time,
+    nested_compare,
+    nested_flatten,
+    nested_map,
+    nested_pack,
+)
+
+
+class TestSerializer:
+    @pytest.mark.parametrize("dtype", [torch.float32, torch.float16, torch.bfloat16, torch.int64, torch.bool])
+    def test_tensor_roundtrip(self, dtype):
+        t = (torch.randn(4, 5) * 10).to(dtype)
+        out = PytorchSerializer.loads(PytorchSerializer.dumps(t))
+        assert out.dtype == dtype and out.shape == t.shape
+        assert torch.equal(out.float(), t.float())
+
+    def test_requires_grad_preserved(self):
+        t = torch.randn(3, requires_grad=True)
+        out = PytorchSerializer.loads(PytorchSerializer.dumps(t))
+        assert out.requires_grad
+
+    def test_nested_structures(self):
+        obj = {"a": [torch.ones(2), {"b": torch.zeros(1, 3)}], "c": "hello", "d": 3.5}
+        out = PytorchSerializer.loads(PytorchSerializer.dumps(obj))
+        assert torch.equal(out["a"][0], torch.ones(2))
+        assert torch.equal(out["a"][1]["b"], torch.zeros(1, 3))
+        assert out["c"] == "hello" and out["d"] == 3.5
+
+    def test_msgpack_plain(self):
+        obj = {"k": [1, 2, "three"], "n": None}
+        assert MSGPackSerializer.loads(MSGPackSerializer.dumps(obj)) == obj
+
+    def test_empty_tensor(self):
+        t = torch.empty(0, 5)
+        out = PytorchSerializer.loads(PytorchSerializer.dumps(t))
+        assert out.shape == (0, 5)
+
+
+class TestNested:
+    def test_flatten_pack_roundtrip(self):
+        s = {"x": (1, [2, 3]), "y": {"z": 4}}
+        flat = list(nested_flatten(s))
+        assert flat == [1, 2, 3, 4]
+        assert nested_pack(flat, s) == s
+
+    def test_map(self):
+        s = (1, {"a": 2})
+        assert nested_map(lambda v: v * 2, s) == (2, {"a": 4})
+
+    def test_map_multi(self):
+        assert nested_map(lambda a, b: a + b, (1, 2), (10, 20)) == (11, 22)
+
+    def test_compare(self):
+        assert nested_compare({"a": (1, 2)}, {"a": (9, 9)})
+        assert not nested_compare({"a": (1, 2)}, {"a": (1, 2, 3)})
+
+    def test_pack_wrong_arity(self):
+        with pytest.raises(ValueError):
+            nested_pack([1, 2, 3], (1, 2))
+
+
+class TestMPFuture:
+    def test_inprocess(self):
+        f = MPFuture()
+        threading.Thread(target=lambda: f.set_result(42)).start()
+        assert f.result(timeout=2.0) == 42
+        assert f.done()
+
+    def test_exception(self):
+        f = MPFuture()
+        f.set_exception(ValueError("boom"))
+        with pytest.raises(ValueError):
+            f.result(0.1)
+
+    def test_timeout(self):
+        f = MPFuture()
+        with pytest.raises(FutureTimeoutError):
+            f.result(timeout=0.05)
+
+    def test_cancel(self):
+        f = MPFuture()
+        assert f.cancel()
+        assert f.cancelled()
+
+    def test_pair_across_process(self):
+        import multiprocessing as mp
+
+        setter, awaiter = MPFuture.make_pair()
+
+        def child(s):
+            s.set_result(torch.ones(3))
+
+        p = mp.Process(target=child, args=(setter,))
+        p.start()
+        result = awaiter.result(timeout=10.0)
+        assert torch.equal(result, torch.ones(3))
+        p.join()
+
+
+class TestTimedStorage:
+    def test_store_get(self):
+        s = TimedStorage()
+        assert s.store("k", "v", get_dht_time() + 10)
+        value, exp = s.get("k")
+        assert value == "v" and exp > get_dht_time()
+
+    def test_expiration(self):
+        s = TimedStorage()
+        s.store("k", "v", get_dht_time() + 0.1)
+        time.sleep(0.15)
+        assert s.get("k") == (None, None)
+
+    def test_fresher_wins(self):
+        s = TimedStorage()
+        t0 = get_dht_time()
+        s.store("k", "old", t0 + 10)
+        assert not s.store("k", "older", t0 + 5)  # staler value rejected
+        assert s.store("k", "new", t0 + 20)
+        assert s.get("k")[0] == "new"
+
+    def test_maxsize_eviction(self):
+        s = TimedStorage(maxsize=2)
+        t0 = get_dht_time()
+        s.store("a", 1, t0 + 5)
+        s.store("b", 2, t0 + 10)
+        s.store("c", 3, t0 + 15)
+        assert len(s) == 2
+        assert s.get("a") == (None, None)  # soonest-to-expire evicted
