@@ -1,0 +1,192 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: training tokens/sec of Transformer-DMoE 256-expert on
+1..8 MI355X GPUs (BASELINE.json headline metric, config #2).
+
+One process per GPU over RCCL; expert-parallel MoE FFN layers (RCCL
+all-to-all dispatch, CDNA4 HIP kernels) + data-parallel trunk with
+asynchronous SGD and periodic Moshpit-style averaging (RcclAverager on a
+side stream) — the reference's no-global-barrier training semantics.
+
+Contract (round driver): `python bench.py --gpus N --steps K --warmup W`;
+for N>1 launched via torch.distributed.run. Rank 0 prints ONE JSON line.
+Synthetic data (no network), random-init weights, bf16 compute.
+"""
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+from learning_at_home_amd.models import TransformerDMoE, TransformerDMoEConfig
+from learning_at_home_amd.parallel import ExpertParallelMoE, FusedExpertSGD
+from learning_at_home_amd.averaging import RcclAverager
+from learning_at_home_amd.ops import HAS_HIP
+
+
+def build_model(cfg: TransformerDMoEConfig, args, device, group=None):
+    def moe_factory(layer_idx, c):
+        return ExpertParallelMoE(
+            d_model=c.d_model, ffn_dim=c.ffn_dim, num_experts=c.num_experts,
+            k=c.k_best, capacity_factor=args.capacity_factor,
+            aux_loss_coef=args.aux_coef, group=group,
+        )
+
+    model = TransformerDMoE(cfg, moe_factory)
+    if device.type == "cuda":
+        model = model.to(device=device, dtype=torch.bfloat16)
+        # fp32 masters for expert stacks + LN in fp32 for stability
+        for block in model.blocks:
+            block.moe.experts.float()
+            block.moe.experts.init_gpu_state()
+    else:
+        model = model.to(device)
+    return model
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--d-model", type=int, default=1024)
+    p.add_argument("--ffn-dim", type=int, default=4096)
+    p.add_argument("--layers", type=int, default=6)
+    p.add_argument("--heads", type=int, default=16)
+    p.add_argument("--experts", type=int, default=256)
+    p.add_argument("--k", type=int, default=4)
+    p.add_argument("--seq", type=int, default=1024)
+    p.add_argument("--batch", type=int, default=8, help="per-GPU batch size")
+    p.add_argument("--vocab", type=int, default=32000)
+    p.add_argument("--capacity-factor", type=float, default=1.5)
+    p.add_argument("--aux-coef", type=float, default=0.01)
+    p.add_argument("--lr", type=float, default=0.01)
+    p.add_argument("--averaging-period", type=int, default=10)
+    p.add_argument("--no-averager", action="store_true")
+    args = p.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    use_cuda = torch.cuda.is_available()
+
+    if world_size > 1:
+        backend = "nccl" if use_cuda else "gloo"
+        if use_cuda:
+            torch.cuda.set_device(local_rank)
+        dist.init_process_group(backend, rank=rank, world_size=world_size)
+    device = torch.device(f"cuda:{local_rank}" if use_cuda else "cpu")
+
+    if use_cuda:
+        cfg = TransformerDMoEConfig(
+            vocab_size=args.vocab, d_model=args.d_model, n_heads=args.heads,
+            n_layers=args.layers, seq_len=args.seq, num_experts=args.experts,
+            k_best=args.k, ffn_dim=args.ffn_dim,
+        )
+        B = args.batch
+    else:  # CPU smoke fallback: tiny config so plain `python bench.py` runs anywhere
+        cfg = TransformerDMoEConfig(
+            vocab_size=1024, d_model=128, n_heads=4, n_layers=2, seq_len=64,
+            num_experts=8, k_best=2, ffn_dim=256,
+        )
+        B = 2
+
+    torch.manual_seed(1234 + rank)
+    model = build_model(cfg, args, device)
+
+    # optimizers: fused async SGD per expert stack; plain SGD for the trunk
+    expert_opts = []
+    trunk_params = []
+    for name, param in model.named_parameters():
+        if ".moe.experts." not in name:
+            trunk_params.append(param)
+    for block in model.blocks:
+        if device.type == "cuda" and HAS_HIP:
+            expert_opts.append(FusedExpertSGD(block.moe.experts, lr=args.lr))
+        else:
+            expert_opts.append(torch.optim.SGD(block.moe.experts.parameters(), lr=args.lr))
+    trunk_opt = torch.optim.SGD(trunk_params, lr=args.lr)
+
+    averager = None
+    if world_size > 1 and not args.no_averager:
+        averager = RcclAverager(trunk_params, dtype=torch.float32 if device.type != "cuda" else torch.bfloat16)
+
+    tokens_per_step_per_rank = B * cfg.seq_len
+
+    def make_batch():
+        ids = torch.randint(0, cfg.vocab_size, (B, cfg.seq_len + 1), device=device)
+        return ids[:, :-1], ids[:, 1:]
+
+    def train_step(step_idx):
+        inputs, labels = make_batch()
+        loss = model.loss(inputs, labels)
+        trunk_opt.zero_grad(set_to_none=True)
+        loss.backward()
+        for opt in expert_opts:
+            opt.step()  # async per-shard SGD (fused on GPU)
+            if isinstance(opt, torch.optim.SGD):
+                opt.zero_grad(set_to_none=True)
+        if averager is not None:
+            averager.wait()  # previous round must have landed before we step
+        trunk_opt.step()
+        if averager is not None and (step_idx + 1) % args.averaging_period == 0:
+            averager.step(async_op=True)  # overlapped on the side stream
+        return loss
+
+    def barrier_sync():
+        if world_size > 1:
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for i in range(args.warmup):
+        train_step(i)
+    barrier_sync()
+    t0 = time.perf_counter()
+    last_loss = None
+    for i in range(args.steps):
+        last_loss = train_step(args.warmup + i)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks = the slowest rank defines the job's wall time
+    if world_size > 1:
+        t = torch.tensor([elapsed], device=device if use_cuda else "cpu",
+                         dtype=torch.float64 if not use_cuda else torch.float32)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    total_tokens_per_s = tokens_per_step_per_rank * world_size * args.steps / elapsed
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "training tokens/sec, Transformer-DMoE 256-expert",
+            "value": round(total_tokens_per_s, 1),
+            "unit": "tokens/s",
+            "n_gpus": world_size,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_cuda else "fp32",
+            "data": "synthetic (random tokens, random-init weights)",
+            "config": {
+                "model": f"Transformer-DMoE d{cfg.d_model} L{cfg.n_layers} "
+                         f"E{cfg.num_experts} top{cfg.k_best} ffn{cfg.ffn_dim}",
+                "global_batch": B * world_size,
+                "seq_len": cfg.seq_len,
+                "parallelism": f"ep{world_size}+dp{world_size} (async SGD, Moshpit averager)",
+                "last_loss": round(float(last_loss), 4) if last_loss is not None else None,
+            },
+        }))
+
+    if world_size > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

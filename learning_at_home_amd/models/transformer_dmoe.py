@@ -83,6 +83,8 @@ class TransformerDMoE(nn.Module):
             moe_factory = lambda i, cfg: DenseFFN(cfg.d_model, cfg.ffn_dim)  # noqa: E731
         self.embed = nn.Embedding(config.vocab_size, config.d_model)
         self.pos_embed = nn.Embedding(config.seq_len, config.d_model)
+        nn.init.normal_(self.embed.weight, std=0.02)
+        nn.init.normal_(self.pos_embed.weight, std=0.02)
         self.blocks = nn.ModuleList(
             [DMoETransformerBlock(config, moe_factory(i, config)) for i in range(config.n_layers)]
         )

@@ -1,0 +1,243 @@
+"""Autograd-integrated ops over the MI355X kernels, with pure-torch CPU
+fallbacks of IDENTICAL semantics (capacity drops, renormalize-over-survivors,
+recomputed-free fused backward). The CPU path doubles as the numerics
+reference for the GPU tests (tests/test_ops_gpu.py compares kernel outputs
+against these functions in fp32).
+
+GEMM policy: the four activation-side GEMMs of the expert FFN (fwd x2,
+dh/dx) are the hand-written MFMA NT kernels with fused bias/GELU epilogues
+(ffn_gemm.hip); the weight-gradient GEMMs (dW1/dW2) currently go through
+torch.bmm (hipBLASLt — a plain library GEMM, no fusion) pending the TN
+kernel; bias grads via the colsum kernel.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+from . import _C, HAS_HIP
+
+# epilogue ids (ffn_gemm.hip)
+EPI_NONE, EPI_BIAS, EPI_BIAS_GELU, EPI_GELU_GRAD = 0, 1, 2, 3
+
+
+# ---------------------------------------------------------------------------
+# gating: softmax + top-k + (optionally) differentiable full probs
+# ---------------------------------------------------------------------------
+class _GatingTopK(torch.autograd.Function):
+    """GPU fused gating. Outputs (topk_idx int32, topk_w fp32, probs bf16).
+    probs is differentiable for TOKEN-UNIFORM upstream grads only (the
+    importance/load-balance losses) — d_probs is reduced to a per-expert
+    vector for the fused backward kernel."""
+
+    @staticmethod
+    def forward(ctx, logits: torch.Tensor, k: int):
+        probs, topk_idx, topk_w = _C.gating_topk_fwd(logits.contiguous(), k)
+        ctx.save_for_backward(probs, topk_idx, topk_w)
+        ctx.k = k
+        ctx.mark_non_differentiable(topk_idx)
+        return topk_idx, topk_w, probs
+
+    @staticmethod
+    def backward(ctx, d_idx, d_topk_w, d_probs):
+        probs, topk_idx, topk_w = ctx.saved_tensors
+        aux_g = None
+        if d_probs is not None:
+            aux_g = d_probs.sum(dim=0).float().contiguous()  # token-uniform reduction
+        if d_topk_w is None:
+            d_topk_w = torch.zeros_like(topk_w)
+        d_logits = _C.gating_topk_bwd(probs, topk_idx, topk_w,
+                                      d_topk_w.float().contiguous(), aux_g)
+        return d_logits, None
+
+
+def gating_topk(logits: torch.Tensor, k: int):
+    """Returns (topk_idx [T,k] int32, topk_w [T,k] fp32 = softmax over the
+    selected logits, probs [T,E] = full softmax)."""
+    if logits.is_cuda and HAS_HIP:
+        return _GatingTopK.apply(logits.float(), k)
+    probs = F.softmax(logits.float(), dim=-1)
+    top_logits, topk_idx = logits.float().topk(k, dim=-1)
+    topk_w = F.softmax(top_logits, dim=-1)
+    return topk_idx.int(), topk_w, probs
+
+
+# ---------------------------------------------------------------------------
+# slot assignment (non-differentiable routing bookkeeping)
+# ---------------------------------------------------------------------------
+def assign_slots(topk_idx: torch.Tensor, E: int, C: int):
+    """Returns (pos [T,k] int32 slot-in-expert or -1 dropped,
+    counts [E] int32, inv [E,C] int32 flat (t*k+j) or -1)."""
+    if topk_idx.is_cuda and HAS_HIP:
+        pos, counts, inv = _C.assign_slots(topk_idx.contiguous(), E, C)
+        return pos, counts, inv
+    T, k = topk_idx.shape
+    flat_e = topk_idx.reshape(-1).long()
+    order = torch.argsort(flat_e, stable=True)
+    sorted_e = flat_e[order]
+    counts_full = torch.bincount(flat_e, minlength=E).int()
+    starts = torch.zeros(E, dtype=torch.long)
+    starts[1:] = torch.cumsum(counts_full, 0)[:-1].long()
+    rank = torch.arange(flat_e.numel()) - starts[sorted_e]
+    pos = torch.full_like(flat_e, -1, dtype=torch.int32)
+    keep = rank < C
+    pos[order[keep]] = rank[keep].int()
+    inv = torch.full((E, C), -1, dtype=torch.int32)
+    kept_slots = sorted_e[keep] * C + rank[keep]
+    inv.view(-1)[kept_slots] = order[keep].int()
+    counts = counts_full
+    return pos.view(T, k), counts, inv
+
+
+# ---------------------------------------------------------------------------
+# dispatch: tokens -> capacity-bucketed expert batches
+# ---------------------------------------------------------------------------
+class _Dispatch(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, topk_idx, pos, inv, k: int):
+        ctx.save_for_backward(topk_idx, pos)
+        ctx.T = x.shape[0]
+        if x.is_cuda and HAS_HIP:
+            return _C.dispatch_gather(x.contiguous(), inv.contiguous(), k)
+        E, C = inv.shape
+        D = x.shape[1]
+        xbuf = torch.zeros(E, C, D, dtype=x.dtype)
+        mask = inv.view(-1) >= 0
+        tok = (inv.view(-1)[mask].long() // k)
+        xbuf.view(-1, D)[mask] = x[tok]
+        return xbuf
+
+    @staticmethod
+    def backward(ctx, dxbuf: torch.Tensor):
+        topk_idx, pos = ctx.saved_tensors
+        T = ctx.T
+        if dxbuf.is_cuda and HAS_HIP:
+            dx = _C.dispatch_bwd(dxbuf.contiguous(), topk_idx, pos, T)
+            return dx, None, None, None, None
+        E, C, D = dxbuf.shape
+        dx = torch.zeros(T, D, dtype=dxbuf.dtype)
+        flat_pos = pos.view(-1).long()
+        flat_e = topk_idx.view(-1).long()
+        keep = flat_pos >= 0
+        rows = dxbuf.view(-1, D)[(flat_e[keep] * C + flat_pos[keep])]
+        tok = torch.div(torch.nonzero(keep).squeeze(1), pos.shape[1], rounding_mode="floor")
+        dx.index_add_(0, tok, rows)
+        return dx, None, None, None, None
+
+
+def dispatch(x, topk_idx, pos, inv, k):
+    return _Dispatch.apply(x, topk_idx, pos, inv, k)
+
+
+# ---------------------------------------------------------------------------
+# combine: weighted sum over surviving slots (renormalized)
+# ---------------------------------------------------------------------------
+class _Combine(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, ybuf: torch.Tensor, topk_idx, pos, w, inv):
+        if ybuf.is_cuda and HAS_HIP:
+            y = _C.combine_fwd(ybuf.contiguous(), topk_idx, pos, w.contiguous())
+        else:
+            y = _combine_fwd_torch(ybuf, topk_idx, pos, w)
+        ctx.save_for_backward(ybuf, y, topk_idx, pos, w, inv)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        ybuf, y, topk_idx, pos, w, inv = ctx.saved_tensors
+        if dy.is_cuda and HAS_HIP:
+            dybuf, dw = _C.combine_bwd(dy.contiguous(), ybuf, y, topk_idx, pos,
+                                       w.contiguous(), inv)
+            return dybuf, None, None, dw, None
+        dybuf, dw = _combine_bwd_torch(dy, ybuf, y, topk_idx, pos, w)
+        return dybuf, None, None, dw, None
+
+
+def _survivor_weights(pos, w):
+    surv = pos >= 0
+    s = (w * surv).sum(-1, keepdim=True)
+    inv_s = torch.where(s > 0, 1.0 / s, torch.zeros_like(s))
+    return (w * surv) * inv_s, surv, inv_s
+
+
+def _combine_fwd_torch(ybuf, topk_idx, pos, w):
+    E, C, D = ybuf.shape
+    T, k = topk_idx.shape
+    w_hat, surv, _ = _survivor_weights(pos, w)
+    flat = (topk_idx.long() * C + pos.clamp(min=0).long()).view(-1)
+    rows = ybuf.view(-1, D)[flat].view(T, k, D)
+    return (w_hat.unsqueeze(-1).to(rows.dtype) * rows * surv.unsqueeze(-1)).sum(1)
+
+
+def _combine_bwd_torch(dy, ybuf, y, topk_idx, pos, w):
+    E, C, D = ybuf.shape
+    T, k = topk_idx.shape
+    w_hat, surv, inv_s = _survivor_weights(pos, w)
+    dybuf = torch.zeros_like(ybuf)
+    flat_valid = pos.view(-1) >= 0
+    flat_slot = (topk_idx.long() * C + pos.clamp(min=0).long()).view(-1)[flat_valid]
+    scaled = (w_hat.unsqueeze(-1).to(dy.dtype) * dy.unsqueeze(1)).view(-1, D)[flat_valid]
+    dybuf.view(-1, D)[flat_slot] = scaled
+    # dw_j = (<dy, y_j> - <dy, y>) / s for survivors
+    rows = ybuf.view(-1, D)[(topk_idx.long() * C + pos.clamp(min=0).long()).view(-1)].view(T, k, D)
+    dot_yj = torch.einsum("td,tkd->tk", dy.float(), rows.float())
+    dot_y = torch.einsum("td,td->t", dy.float(), y.float()).unsqueeze(-1)
+    dw = ((dot_yj - dot_y) * inv_s) * surv
+    return dybuf, dw.float()
+
+
+def combine(ybuf, topk_idx, pos, w, inv):
+    return _Combine.apply(ybuf, topk_idx, pos, w, inv)
+
+
+# ---------------------------------------------------------------------------
+# batched expert FFN (GPU fused path; weight grads -> stack.main_grad_*)
+# ---------------------------------------------------------------------------
+class _BatchedFFNHip(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, xbuf: torch.Tensor, stack):
+        E, C, D = xbuf.shape
+        pre1 = torch.empty(E, C, stack.ffn_dim, dtype=torch.bfloat16, device=xbuf.device)
+        h = _C.batched_gemm_nt(xbuf.contiguous(), stack.W1c, stack.b1_f32, pre1, EPI_BIAS_GELU)
+        y = _C.batched_gemm_nt(h, stack.W2c, stack.b2_f32, None, EPI_BIAS)
+        ctx.save_for_backward(xbuf, pre1, h)
+        ctx.stack = stack
+        return y
+
+    @staticmethod
+    def backward(ctx, dy: torch.Tensor):
+        xbuf, pre1, h = ctx.saved_tensors
+        stack = ctx.stack
+        dy = dy.contiguous()
+        # dh_pre = (dy @ W2t^T) * gelu'(pre1); dx = dh_pre @ W1t^T   (MFMA NT)
+        dh_pre = _C.batched_gemm_nt(dy, stack.W2t, None, pre1, EPI_GELU_GRAD)
+        dxbuf = _C.batched_gemm_nt(dh_pre, stack.W1t, None, None, EPI_NONE)
+        # weight grads: plain library batched GEMMs (hipBLASLt) + colsum kernel
+        stack.main_grad_W1.add_(torch.bmm(dh_pre.transpose(1, 2), xbuf).float())
+        stack.main_grad_W2.add_(torch.bmm(dy.transpose(1, 2), h).float())
+        stack.main_grad_b1.add_(_C.colsum(dh_pre))
+        stack.main_grad_b2.add_(_C.colsum(dy))
+        stack.grad_ready = True
+        return dxbuf, None
+
+
+def batched_ffn(xbuf: torch.Tensor, stack) -> torch.Tensor:
+    """xbuf [E, C, D] -> y [E, C, D] through each expert's FFN."""
+    if xbuf.is_cuda and HAS_HIP:
+        return _BatchedFFNHip.apply(xbuf, stack)
+    # CPU fallback: differentiable torch path over the fp32 master weights
+    h_pre = torch.einsum("ecd,efd->ecf", xbuf.float(), stack.W1.float()) + stack.W1_bias.float()[:, None, :]
+    h = F.gelu(h_pre, approximate="tanh")
+    y = torch.einsum("ecf,edf->ecd", h, stack.W2.float()) + stack.W2_bias.float()[:, None, :]
+    return y.to(xbuf.dtype)
+
+
+# ---------------------------------------------------------------------------
+# numerics references (fp32, used by GPU kernel tests)
+# ---------------------------------------------------------------------------
+def expert_ffn_reference(x, W1, b1, W2, b2):
+    """fp32 reference of one expert FFN: gelu_tanh(x@W1^T+b1)@W2^T+b2."""
+    h = F.gelu(x.float() @ W1.float().t() + b1.float(), approximate="tanh")
+    return h @ W2.float().t() + b2.float()

@@ -1,0 +1,196 @@
+// Python bindings for the MI355X kernel suite (torch extension).
+// Thin layer: dtype/contiguity checks, output allocation, stream plumbing.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void lah_batched_gemm_nt(const void*, const void*, void*, const void*, void*, int, int,
+                         int, int, int, hipStream_t);
+void lah_gating_topk_fwd(const void*, void*, void*, void*, int, int, int, hipStream_t);
+void lah_gating_topk_bwd(const void*, const void*, const void*, const void*, const void*,
+                         void*, int, int, int, hipStream_t);
+void lah_assign_slots(const void*, void*, void*, void*, int, int, int, hipStream_t);
+void lah_dispatch_gather(const void*, const void*, void*, int, int, int, int, hipStream_t);
+void lah_dispatch_bwd(const void*, const void*, const void*, void*, int, int, int, int,
+                      hipStream_t);
+void lah_combine_fwd(const void*, const void*, const void*, const void*, void*, int, int,
+                     int, int, hipStream_t);
+void lah_combine_bwd(const void*, const void*, const void*, const void*, const void*,
+                     const void*, void*, void*, int, int, int, int, hipStream_t);
+void lah_zero_padding(const void*, void*, int, int, int, hipStream_t);
+void lah_sgd_step(void*, const void*, void*, void*, long, float, float, float,
+                  hipStream_t);
+void lah_transpose_bf16(const void*, void*, int, int, int, hipStream_t);
+void lah_colsum(const void*, void*, int, int, int, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() { return at::hip::getCurrentHIPStream().stream(); }
+
+#define CHECK_BF16(t) TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16")
+#define CHECK_F32(t) TORCH_CHECK((t).scalar_type() == at::kFloat, #t " must be fp32")
+#define CHECK_I32(t) TORCH_CHECK((t).scalar_type() == at::kInt, #t " must be int32")
+#define CHECK_CONTIG(t) TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+#define CHECK_CUDA(t) TORCH_CHECK((t).is_cuda(), #t " must be on GPU")
+
+// C[e] = A[e] @ B[e]^T with fused epilogue; see ffn_gemm.hip
+torch::Tensor batched_gemm_nt(torch::Tensor A, torch::Tensor B,
+                              c10::optional<torch::Tensor> bias,
+                              c10::optional<torch::Tensor> pre, int64_t epi) {
+  CHECK_CUDA(A); CHECK_BF16(A); CHECK_CONTIG(A);
+  CHECK_BF16(B); CHECK_CONTIG(B);
+  int E = A.size(0), M = A.size(1), K = A.size(2);
+  int N = B.size(1);
+  TORCH_CHECK(B.size(0) == E && B.size(2) == K, "A/B shape mismatch");
+  TORCH_CHECK(N % 128 == 0, "N must be a multiple of 128");
+  TORCH_CHECK(K % 64 == 0, "K must be a multiple of 64");
+  auto C = torch::empty({E, M, N}, A.options());
+  const void* bias_p = nullptr;
+  void* pre_p = nullptr;
+  if (bias.has_value()) { CHECK_F32(*bias); CHECK_CONTIG(*bias); bias_p = bias->data_ptr(); }
+  if (epi == 2) {  // BIAS_GELU stores pre-activation
+    TORCH_CHECK(pre.has_value(), "epi=2 requires a pre-activation output tensor");
+  }
+  if (pre.has_value()) { CHECK_BF16(*pre); CHECK_CONTIG(*pre); pre_p = pre->data_ptr(); }
+  lah_batched_gemm_nt(A.data_ptr(), B.data_ptr(), C.data_ptr(), bias_p, pre_p, E, M, N, K,
+                      (int)epi, cur_stream());
+  return C;
+}
+
+std::vector<torch::Tensor> gating_topk_fwd(torch::Tensor logits, int64_t k) {
+  CHECK_CUDA(logits); CHECK_F32(logits); CHECK_CONTIG(logits);
+  int T = logits.size(0), E = logits.size(1);
+  auto probs = torch::empty({T, E}, logits.options().dtype(at::kBFloat16));
+  auto topk_idx = torch::empty({T, k}, logits.options().dtype(at::kInt));
+  auto topk_w = torch::empty({T, k}, logits.options());
+  lah_gating_topk_fwd(logits.data_ptr(), probs.data_ptr(), topk_idx.data_ptr(),
+                      topk_w.data_ptr(), T, E, (int)k, cur_stream());
+  return {probs, topk_idx, topk_w};
+}
+
+torch::Tensor gating_topk_bwd(torch::Tensor probs, torch::Tensor topk_idx,
+                              torch::Tensor topk_w, torch::Tensor d_topk_w,
+                              c10::optional<torch::Tensor> aux_g) {
+  CHECK_BF16(probs); CHECK_I32(topk_idx); CHECK_F32(topk_w); CHECK_F32(d_topk_w);
+  int T = probs.size(0), E = probs.size(1), k = topk_idx.size(1);
+  auto d_logits = torch::empty({T, E}, topk_w.options());
+  const void* aux_p = nullptr;
+  if (aux_g.has_value()) { CHECK_F32(*aux_g); CHECK_CONTIG(*aux_g); aux_p = aux_g->data_ptr(); }
+  lah_gating_topk_bwd(probs.data_ptr(), topk_idx.data_ptr(), topk_w.data_ptr(),
+                      d_topk_w.contiguous().data_ptr(), aux_p, d_logits.data_ptr(), T, E, k,
+                      cur_stream());
+  return d_logits;
+}
+
+std::vector<torch::Tensor> assign_slots(torch::Tensor topk_idx, int64_t E, int64_t C) {
+  CHECK_CUDA(topk_idx); CHECK_I32(topk_idx); CHECK_CONTIG(topk_idx);
+  int T = topk_idx.size(0), k = topk_idx.size(1);
+  auto pos = torch::empty({T, k}, topk_idx.options());
+  auto counts = torch::zeros({E}, topk_idx.options());
+  auto inv = torch::full({E, C}, -1, topk_idx.options());
+  lah_assign_slots(topk_idx.data_ptr(), pos.data_ptr(), counts.data_ptr(), inv.data_ptr(),
+                   T, k, (int)C, cur_stream());
+  return {pos, counts, inv};
+}
+
+torch::Tensor dispatch_gather(torch::Tensor x, torch::Tensor inv, int64_t k) {
+  CHECK_BF16(x); CHECK_CONTIG(x); CHECK_I32(inv); CHECK_CONTIG(inv);
+  int E = inv.size(0), C = inv.size(1), D = x.size(1);
+  TORCH_CHECK(D % 8 == 0, "D must be a multiple of 8");
+  auto xbuf = torch::empty({E, C, D}, x.options());
+  lah_dispatch_gather(x.data_ptr(), inv.data_ptr(), xbuf.data_ptr(), E, C, D, (int)k,
+                      cur_stream());
+  return xbuf;
+}
+
+torch::Tensor dispatch_bwd(torch::Tensor dxbuf, torch::Tensor topk_idx, torch::Tensor pos,
+                           int64_t T) {
+  CHECK_BF16(dxbuf); CHECK_CONTIG(dxbuf);
+  int C = dxbuf.size(1), D = dxbuf.size(2), k = topk_idx.size(1);
+  auto dx = torch::empty({T, D}, dxbuf.options());
+  lah_dispatch_bwd(dxbuf.data_ptr(), topk_idx.data_ptr(), pos.data_ptr(), dx.data_ptr(),
+                   (int)T, D, k, C, cur_stream());
+  return dx;
+}
+
+torch::Tensor combine_fwd(torch::Tensor ybuf, torch::Tensor topk_idx, torch::Tensor pos,
+                          torch::Tensor w) {
+  CHECK_BF16(ybuf); CHECK_CONTIG(ybuf); CHECK_F32(w);
+  int C = ybuf.size(1), D = ybuf.size(2);
+  int T = topk_idx.size(0), k = topk_idx.size(1);
+  auto y = torch::empty({T, D}, ybuf.options());
+  lah_combine_fwd(ybuf.data_ptr(), topk_idx.data_ptr(), pos.data_ptr(), w.data_ptr(),
+                  y.data_ptr(), T, D, k, C, cur_stream());
+  return y;
+}
+
+std::vector<torch::Tensor> combine_bwd(torch::Tensor dy, torch::Tensor ybuf,
+                                       torch::Tensor y, torch::Tensor topk_idx,
+                                       torch::Tensor pos, torch::Tensor w,
+                                       torch::Tensor inv) {
+  CHECK_BF16(dy); CHECK_BF16(ybuf); CHECK_BF16(y);
+  int E = ybuf.size(0), C = ybuf.size(1), D = ybuf.size(2);
+  int T = topk_idx.size(0), k = topk_idx.size(1);
+  auto dybuf = torch::empty({E, C, D}, ybuf.options());
+  auto dw = torch::empty({T, k}, w.options());
+  // padding rows (inv < 0) are zero-filled; surviving rows written by scatter
+  lah_zero_padding(inv.data_ptr(), dybuf.data_ptr(), E, C, D, cur_stream());
+  lah_combine_bwd(dy.contiguous().data_ptr(), ybuf.data_ptr(), y.data_ptr(),
+                  topk_idx.data_ptr(), pos.data_ptr(), w.data_ptr(), dybuf.data_ptr(),
+                  dw.data_ptr(), T, D, k, C, cur_stream());
+  return {dybuf, dw};
+}
+
+void sgd_step(torch::Tensor master, torch::Tensor grad, c10::optional<torch::Tensor> momentum,
+              torch::Tensor out_bf16, double lr, double mu, double weight_decay) {
+  CHECK_F32(master); CHECK_F32(grad); CHECK_BF16(out_bf16);
+  CHECK_CONTIG(master); CHECK_CONTIG(grad); CHECK_CONTIG(out_bf16);
+  long n = master.numel();
+  TORCH_CHECK(grad.numel() == n && out_bf16.numel() == n, "size mismatch");
+  TORCH_CHECK(n % 4 == 0, "numel must be a multiple of 4");
+  void* mom_p = nullptr;
+  if (momentum.has_value()) { CHECK_F32(*momentum); mom_p = momentum->data_ptr(); }
+  lah_sgd_step(master.data_ptr(), grad.data_ptr(), mom_p, out_bf16.data_ptr(), n, (float)lr,
+               (float)mu, (float)weight_decay, cur_stream());
+}
+
+torch::Tensor transpose_bf16(torch::Tensor in) {
+  CHECK_BF16(in); CHECK_CONTIG(in); CHECK_CUDA(in);
+  int E = in.size(0), R = in.size(1), C = in.size(2);
+  TORCH_CHECK(R % 64 == 0 && C % 64 == 0, "transpose dims must be multiples of 64");
+  auto out = torch::empty({E, C, R}, in.options());
+  lah_transpose_bf16(in.data_ptr(), out.data_ptr(), E, R, C, cur_stream());
+  return out;
+}
+
+torch::Tensor colsum(torch::Tensor d) {
+  CHECK_BF16(d); CHECK_CONTIG(d);
+  int E = d.size(0), M = d.size(1), C = d.size(2);
+  auto db = torch::empty({E, C}, d.options().dtype(at::kFloat));
+  lah_colsum(d.data_ptr(), db.data_ptr(), E, M, C, cur_stream());
+  return db;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("batched_gemm_nt", &batched_gemm_nt, "batched NT GEMM with fused epilogue",
+        py::arg("A"), py::arg("B"), py::arg("bias") = py::none(),
+        py::arg("pre") = py::none(), py::arg("epi") = 0);
+  m.def("gating_topk_fwd", &gating_topk_fwd);
+  m.def("gating_topk_bwd", &gating_topk_bwd, py::arg("probs"), py::arg("topk_idx"),
+        py::arg("topk_w"), py::arg("d_topk_w"), py::arg("aux_g") = py::none());
+  m.def("assign_slots", &assign_slots);
+  m.def("dispatch_gather", &dispatch_gather);
+  m.def("dispatch_bwd", &dispatch_bwd);
+  m.def("combine_fwd", &combine_fwd);
+  m.def("combine_bwd", &combine_bwd);
+  m.def("sgd_step", &sgd_step, py::arg("master"), py::arg("grad"),
+        py::arg("momentum"), py::arg("out_bf16"),
+        py::arg("lr") = 0.01, py::arg("mu") = 0.0, py::arg("weight_decay") = 0.0);
+  m.def("transpose_bf16", &transpose_bf16);
+  m.def("colsum", &colsum);
+}

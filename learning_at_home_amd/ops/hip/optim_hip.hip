@@ -1,0 +1,105 @@
+#include "hip/hip_runtime.h"
+// Fused asynchronous-SGD step for expert shards (SURVEY.md §2 hot-op #5:
+// per-expert optimizer fused into one pass) + batched bf16 transpose that
+// maintains the W^T copies the NT-only GEMM plan needs (ffn_gemm.hip).
+//
+// One pass per expert-stack: read fp32 master + fp32 grad, update master
+// (optional momentum / weight decay), write fp32 master and the bf16 compute
+// copy. Memory-bound by design: everything is 16-byte vectorized.
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(4))) float f4;
+
+__global__ void sgd_step_kernel(float* __restrict__ master, const float* __restrict__ grad,
+                                float* __restrict__ momentum,  // null = plain SGD
+                                __bf16* __restrict__ out_bf16, long n, float lr,
+                                float mu, float weight_decay) {
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
+  long stride = (long)gridDim.x * blockDim.x * 4;
+  for (; i + 3 < n; i += stride) {
+    f4 m = *(const f4*)(master + i);
+    f4 g = *(const f4*)(grad + i);
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      float gq = g[q] + weight_decay * m[q];
+      if (momentum != nullptr) {
+        float v = momentum[i + q] * mu + gq;
+        momentum[i + q] = v;
+        gq = v;
+      }
+      m[q] -= lr * gq;
+    }
+    *(f4*)(master + i) = m;
+    bf16x4 o;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) o[q] = f2bf(m[q]);
+    *(bf16x4*)(out_bf16 + i) = o;
+  }
+}
+
+// batched transpose: out[e][c][r] = in[e][r][c], bf16, 64x64 LDS tiles
+// (+8-element row pad so read-back columns hit distinct banks)
+#define TP 64
+__global__ void transpose_bf16_kernel(const __bf16* __restrict__ in, __bf16* __restrict__ out,
+                                      int R, int C) {
+  __shared__ __bf16 tile[TP][TP + 8];
+  int e = blockIdx.z;
+  int r0 = blockIdx.y * TP, c0 = blockIdx.x * TP;
+  const __bf16* ine = in + (long)e * R * C;
+  __bf16* oute = out + (long)e * R * C;
+  // 256 threads: each loads 16 rows x 8 cols (bf16x8)
+  int tr = threadIdx.x >> 3;          // 0..31  (row within tile, x2)
+  int tc = (threadIdx.x & 7) * 8;     // col within tile (8-elem chunks)
+#pragma unroll
+  for (int s = 0; s < TP; s += 32) {
+    int r = tr + s;
+    bf16x8 v = *(const bf16x8*)(ine + (long)(r0 + r) * C + c0 + tc);
+#pragma unroll
+    for (int q = 0; q < 8; ++q) tile[tc + q][r] = v[q];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int s = 0; s < TP; s += 32) {
+    int r = tr + s;  // now a row of the TRANSPOSED tile (original column)
+    bf16x8 v;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) v[q] = tile[r][tc + q];
+    *(bf16x8*)(oute + (long)(c0 + r) * R + r0 + tc) = v;
+  }
+}
+
+// column sum for bias grads: db[e][c] = sum_m d[e][m][c], fp32 out
+__global__ void colsum_kernel(const __bf16* __restrict__ d, float* __restrict__ db,
+                              int M, int C) {
+  int e = blockIdx.z;
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  const __bf16* de = d + (long)e * M * C;
+  float acc = 0.f;
+  for (int m = 0; m < M; ++m) acc += bf2f(de[(long)m * C + c]);
+  db[(long)e * C + c] = acc;
+}
+
+extern "C" void lah_sgd_step(void* master, const void* grad, void* momentum,
+                             void* out_bf16, long n, float lr, float mu,
+                             float weight_decay, hipStream_t stream) {
+  long blocks = (n / 4 + 255) / 256;
+  if (blocks > 2048) blocks = 2048;  // grid-stride (guide G11)
+  hipLaunchKernelGGL(sgd_step_kernel, dim3((unsigned)blocks), dim3(256), 0, stream,
+                     (float*)master, (const float*)grad, (float*)momentum,
+                     (__bf16*)out_bf16, n, lr, mu, weight_decay);
+}
+
+extern "C" void lah_transpose_bf16(const void* in, void* out, int E, int R, int C,
+                                   hipStream_t stream) {
+  dim3 grid(C / TP, R / TP, E), block(256);
+  hipLaunchKernelGGL(transpose_bf16_kernel, grid, block, 0, stream, (const __bf16*)in,
+                     (__bf16*)out, R, C);
+}
+
+extern "C" void lah_colsum(const void* d, void* db, int E, int M, int C,
+                           hipStream_t stream) {
+  dim3 grid(cdiv(C, 256), 1, E), block(256);
+  hipLaunchKernelGGL(colsum_kernel, grid, block, 0, stream, (const __bf16*)d, (float*)db,
+                     M, C);
+}

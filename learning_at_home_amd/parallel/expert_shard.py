@@ -1,0 +1,220 @@
+"""Expert-parallel MoE layer — the MI355X tensor plane (SURVEY.md §2
+"TaskPool/ConnectionHandler MI355X equivalent", §5 "Distributed communication
+backend"): token dispatch and gradient return ride RCCL all-to-all over xGMI
+between one-process-per-GPU ranks; each rank owns an ExpertStack shard whose
+FFNs run through the hand-written CDNA4 kernels (ops/).
+
+Fault-tolerance semantics preserved from the reference: slots over a fixed
+per-expert CAPACITY are dropped and the combine renormalizes the gating
+softmax over the surviving slots (= "drop non-responders, renormalize"),
+asynchronous SGD per expert shard (no global optimizer barrier).
+
+Memory layout (288 GB HBM3E per GPU): per-layer stacked weights
+W1 [E_local, ffn, d] / W2 [E_local, d, ffn] as fp32 masters + bf16 compute
+copies + bf16 transposed copies (so every activation GEMM is MFMA-natural
+NT — see ops/hip/ffn_gemm.hip) + fp32 main_grad buffers consumed by the
+fused SGD kernel.
+"""
+from __future__ import annotations
+
+import math
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ..ops import HAS_HIP, _C
+from ..ops import functional as Fops
+from ..utils import get_logger
+
+logger = get_logger(__name__)
+
+
+class ExpertStack(nn.Module):
+    """E_local experts' FFN weights, stacked for grouped kernels."""
+
+    def __init__(self, num_local: int, d_model: int, ffn_dim: int):
+        super().__init__()
+        self.num_local, self.d_model, self.ffn_dim = num_local, d_model, ffn_dim
+        # fp32 masters (the checkpointable parameters; nn.Linear-style init)
+        bound1 = 1.0 / math.sqrt(d_model)
+        bound2 = 1.0 / math.sqrt(ffn_dim)
+        self.W1 = nn.Parameter(torch.empty(num_local, ffn_dim, d_model).uniform_(-bound1, bound1))
+        self.W1_bias = nn.Parameter(torch.empty(num_local, ffn_dim).uniform_(-bound1, bound1))
+        self.W2 = nn.Parameter(torch.empty(num_local, d_model, ffn_dim).uniform_(-bound2, bound2))
+        self.W2_bias = nn.Parameter(torch.empty(num_local, d_model).uniform_(-bound2, bound2))
+        self._gpu_state_ready = False
+        self.grad_ready = False
+
+    # ---- GPU compute-state management ---------------------------------
+    def init_gpu_state(self) -> None:
+        """Allocate bf16 compute/transposed copies + fp32 grad buffers."""
+        assert self.W1.is_cuda, "move the stack to GPU first"
+        dev = self.W1.device
+        self.W1c = self.W1.detach().bfloat16().contiguous()
+        self.W2c = self.W2.detach().bfloat16().contiguous()
+        self.W1t = _C.transpose_bf16(self.W1c)  # [E, d, ffn]
+        self.W2t = _C.transpose_bf16(self.W2c)  # [E, ffn, d]
+        self.b1_f32 = self.W1_bias.detach().float().contiguous()
+        self.b2_f32 = self.W2_bias.detach().float().contiguous()
+        self.main_grad_W1 = torch.zeros_like(self.W1, device=dev)
+        self.main_grad_W2 = torch.zeros_like(self.W2, device=dev)
+        self.main_grad_b1 = torch.zeros_like(self.W1_bias, device=dev)
+        self.main_grad_b2 = torch.zeros_like(self.W2_bias, device=dev)
+        self._gpu_state_ready = True
+
+    def zero_main_grads(self) -> None:
+        for g in (self.main_grad_W1, self.main_grad_W2, self.main_grad_b1, self.main_grad_b2):
+            g.zero_()
+        self.grad_ready = False
+
+
+class FusedExpertSGD:
+    """Asynchronous per-shard SGD: one fused kernel pass per tensor updates
+    the fp32 master AND re-quantizes the bf16 compute copy; transposed copies
+    refreshed by the tiled transpose kernel. (SURVEY.md §2 hot-op #5.)"""
+
+    def __init__(self, stack: ExpertStack, lr: float = 0.01, momentum: float = 0.0,
+                 weight_decay: float = 0.0):
+        self.stack = stack
+        self.lr, self.momentum, self.weight_decay = lr, momentum, weight_decay
+        self.momentum_bufs = None
+        if momentum > 0:
+            self.momentum_bufs = [
+                torch.zeros_like(stack.W1), torch.zeros_like(stack.W2),
+                torch.zeros_like(stack.W1_bias), torch.zeros_like(stack.W2_bias),
+            ]
+        self.step_count = 0
+
+    @torch.no_grad()
+    def step(self) -> None:
+        st = self.stack
+        if not st.grad_ready:
+            return
+        mom = self.momentum_bufs or [None] * 4
+        _C.sgd_step(st.W1.data.view(-1), st.main_grad_W1.view(-1),
+                    mom[0].view(-1) if mom[0] is not None else None,
+                    st.W1c.view(-1), self.lr, self.momentum, self.weight_decay)
+        _C.sgd_step(st.W2.data.view(-1), st.main_grad_W2.view(-1),
+                    mom[1].view(-1) if mom[1] is not None else None,
+                    st.W2c.view(-1), self.lr, self.momentum, self.weight_decay)
+        _C.sgd_step(st.W1_bias.data.view(-1), st.main_grad_b1.view(-1),
+                    mom[2].view(-1) if mom[2] is not None else None,
+                    torch.empty_like(st.W1_bias, dtype=torch.bfloat16).view(-1),
+                    self.lr, self.momentum, self.weight_decay)
+        _C.sgd_step(st.W2_bias.data.view(-1), st.main_grad_b2.view(-1),
+                    mom[3].view(-1) if mom[3] is not None else None,
+                    torch.empty_like(st.W2_bias, dtype=torch.bfloat16).view(-1),
+                    self.lr, self.momentum, self.weight_decay)
+        st.b1_f32.copy_(st.W1_bias.data)
+        st.b2_f32.copy_(st.W2_bias.data)
+        st.W1t = _C.transpose_bf16(st.W1c)
+        st.W2t = _C.transpose_bf16(st.W2c)
+        st.zero_main_grads()
+        self.step_count += 1
+
+
+class _AllToAll(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, group):
+        ctx.group = group
+        out = torch.empty_like(x)
+        dist.all_to_all_single(out, x.contiguous(), group=group)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad: torch.Tensor):
+        out = torch.empty_like(grad)
+        dist.all_to_all_single(out, grad.contiguous(), group=ctx.group)
+        return out, None
+
+
+class ExpertParallelMoE(nn.Module):
+    """Drop-in MoE FFN layer: [.., D] -> [.., D], experts sharded over the
+    process group (E % world_size == 0), RCCL all-to-all dispatch/return."""
+
+    def __init__(
+        self,
+        d_model: int,
+        ffn_dim: int,
+        num_experts: int,
+        k: int = 4,
+        capacity_factor: float = 1.25,
+        aux_loss_coef: float = 0.01,
+        group: Optional[object] = None,
+        tokens_per_batch_hint: int = 8192,
+    ):
+        super().__init__()
+        self.d_model, self.ffn_dim = d_model, ffn_dim
+        self.num_experts, self.k = num_experts, k
+        self.capacity_factor = capacity_factor
+        self.aux_loss_coef = aux_loss_coef
+        self.group = group
+        self.world_size = dist.get_world_size(group) if dist.is_initialized() else 1
+        self.rank = dist.get_rank(group) if dist.is_initialized() else 0
+        assert num_experts % self.world_size == 0, "experts must shard evenly"
+        self.num_local = num_experts // self.world_size
+        self.gate = nn.Linear(d_model, num_experts, bias=False)
+        self.experts = ExpertStack(self.num_local, d_model, ffn_dim)
+        self.aux_loss: Optional[torch.Tensor] = None
+        self.last_counts: Optional[torch.Tensor] = None
+        self._capacity_cache: dict = {}
+
+    def capacity(self, T: int) -> int:
+        """Per-(source rank, expert) capacity, padded to the GEMM tile (128
+        on GPU where the MFMA kernel runs, 8 on CPU tests)."""
+        key = T
+        if key not in self._capacity_cache:
+            avg = T * self.k / self.num_experts
+            pad = 64 if (HAS_HIP and self.gate.weight.is_cuda) else 8
+            cap = max(pad, int(math.ceil(avg * self.capacity_factor / pad)) * pad)
+            self._capacity_cache[key] = cap
+        return self._capacity_cache[key]
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        lead_shape = x.shape[:-1]
+        D = x.shape[-1]
+        flat = x.reshape(-1, D)
+        T = flat.shape[0]
+        E, k, W = self.num_experts, self.k, self.world_size
+        C = self.capacity(T)
+
+        logits = self.gate(flat).float()
+        topk_idx, topk_w, probs = Fops.gating_topk(logits, k)
+        with torch.no_grad():
+            pos, counts, inv = Fops.assign_slots(topk_idx, E, C)
+        self.last_counts = counts
+
+        compute_dtype = torch.bfloat16 if flat.is_cuda else flat.dtype
+        xbuf = Fops.dispatch(flat.to(compute_dtype), topk_idx, pos, inv, k)  # [E, C, D]
+
+        if W > 1:
+            xbuf = _AllToAll.apply(xbuf, self.group)  # [W*E_local, C, D] -> recv
+            xrecv = (
+                xbuf.view(W, self.num_local, C, D).permute(1, 0, 2, 3)
+                .reshape(self.num_local, W * C, D).contiguous()
+            )
+        else:
+            xrecv = xbuf
+
+        ybuf_local = Fops.batched_ffn(xrecv, self.experts)
+
+        if W > 1:
+            ysend = (
+                ybuf_local.view(self.num_local, W, C, D).permute(1, 0, 2, 3)
+                .reshape(E, C, D).contiguous()
+            )
+            ybuf = _AllToAll.apply(ysend, self.group)
+        else:
+            ybuf = ybuf_local
+
+        y = Fops.combine(ybuf, topk_idx, pos, topk_w, inv)
+
+        if self.aux_loss_coef > 0:
+            f = (counts.float() / max(1, T * k)).detach()
+            P = probs.float().mean(dim=0)
+            self.aux_loss = self.aux_loss_coef * E * torch.dot(f.to(P.device), P)
+        else:
+            self.aux_loss = None
+        return y.reshape(*lead_shape, D).to(x.dtype)
