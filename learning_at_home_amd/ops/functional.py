@@ -214,11 +214,19 @@ class _BatchedFFNHip(torch.autograd.Function):
         # dh_pre = (dy @ W2t^T) * gelu'(pre1); dx = dh_pre @ W1t^T   (MFMA NT)
         dh_pre = _C.batched_gemm_nt(dy, stack.W2t, None, pre1, EPI_GELU_GRAD)
         dxbuf = _C.batched_gemm_nt(dh_pre, stack.W1t, None, None, EPI_NONE)
-        # weight grads: plain library batched GEMMs (hipBLASLt) + colsum kernel
-        stack.main_grad_W1.add_(torch.bmm(dh_pre.transpose(1, 2), xbuf).float())
-        stack.main_grad_W2.add_(torch.bmm(dy.transpose(1, 2), h).float())
-        stack.main_grad_b1.add_(_C.colsum(dh_pre))
-        stack.main_grad_b2.add_(_C.colsum(dy))
+        # weight grads: hand-written TN MFMA kernels write the fp32 grad
+        # buffers in place (accumulate only after the first backward of a step)
+        acc = stack.grad_ready
+        _C.batched_gemm_tn(dh_pre, xbuf, stack.main_grad_W1, acc)
+        _C.batched_gemm_tn(dy, h, stack.main_grad_W2, acc)
+        db1 = _C.colsum(dh_pre)
+        db2 = _C.colsum(dy)
+        if acc:
+            stack.main_grad_b1.add_(db1)
+            stack.main_grad_b2.add_(db2)
+        else:
+            stack.main_grad_b1.copy_(db1)
+            stack.main_grad_b2.copy_(db2)
         stack.grad_ready = True
         return dxbuf, None
 

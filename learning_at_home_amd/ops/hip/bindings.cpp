@@ -8,6 +8,8 @@
 extern "C" {
 void lah_batched_gemm_nt(const void*, const void*, void*, const void*, void*, int, int,
                          int, int, int, hipStream_t);
+void lah_batched_gemm_tn(const void*, const void*, void*, int, int, int, int, int,
+                         hipStream_t);
 void lah_gating_topk_fwd(const void*, void*, void*, void*, int, int, int, hipStream_t);
 void lah_gating_topk_bwd(const void*, const void*, const void*, const void*, const void*,
                          void*, int, int, int, hipStream_t);
@@ -22,6 +24,8 @@ void lah_combine_bwd(const void*, const void*, const void*, const void*, const v
 void lah_zero_padding(const void*, void*, int, int, int, hipStream_t);
 void lah_sgd_step(void*, const void*, void*, void*, long, float, float, float,
                   hipStream_t);
+void lah_sgd_step_t(void*, const void*, void*, void*, int, int, int, float, float,
+                    hipStream_t);
 void lah_transpose_bf16(const void*, void*, int, int, int, hipStream_t);
 void lah_colsum(const void*, void*, int, int, int, hipStream_t);
 }
@@ -58,6 +62,20 @@ torch::Tensor batched_gemm_nt(torch::Tensor A, torch::Tensor B,
   lah_batched_gemm_nt(A.data_ptr(), B.data_ptr(), C.data_ptr(), bias_p, pre_p, E, M, N, K,
                       (int)epi, cur_stream());
   return C;
+}
+
+// C[e,i,j] (fp32, written in place) = sum_m A[e,m,i] * B[e,m,j]
+void batched_gemm_tn(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+                     bool accumulate) {
+  CHECK_CUDA(A); CHECK_BF16(A); CHECK_CONTIG(A);
+  CHECK_BF16(B); CHECK_CONTIG(B); CHECK_F32(C); CHECK_CONTIG(C);
+  int E = A.size(0), M = A.size(1), I = A.size(2), J = B.size(2);
+  TORCH_CHECK(B.size(0) == E && B.size(1) == M, "A/B shape mismatch");
+  TORCH_CHECK(C.size(0) == E && C.size(1) == I && C.size(2) == J, "C shape mismatch");
+  TORCH_CHECK(M % 64 == 0, "M must be a multiple of 64");
+  TORCH_CHECK(I % 128 == 0 && J % 128 == 0, "I/J must be multiples of 128");
+  lah_batched_gemm_tn(A.data_ptr(), B.data_ptr(), C.data_ptr(), E, M, I, J,
+                      accumulate ? 1 : 0, cur_stream());
 }
 
 std::vector<torch::Tensor> gating_topk_fwd(torch::Tensor logits, int64_t k) {
@@ -157,6 +175,17 @@ void sgd_step(torch::Tensor master, torch::Tensor grad, c10::optional<torch::Ten
                (float)mu, (float)weight_decay, cur_stream());
 }
 
+void sgd_step_t(torch::Tensor master, torch::Tensor grad, torch::Tensor out,
+                torch::Tensor outT, double lr, double weight_decay) {
+  CHECK_F32(master); CHECK_F32(grad); CHECK_BF16(out); CHECK_BF16(outT);
+  CHECK_CONTIG(master); CHECK_CONTIG(grad); CHECK_CONTIG(out); CHECK_CONTIG(outT);
+  int E = master.size(0), R = master.size(1), C = master.size(2);
+  TORCH_CHECK(R % 64 == 0 && C % 64 == 0, "dims must be multiples of 64");
+  TORCH_CHECK(outT.size(1) == C && outT.size(2) == R, "outT must be [E, C, R]");
+  lah_sgd_step_t(master.data_ptr(), grad.data_ptr(), out.data_ptr(), outT.data_ptr(),
+                 E, R, C, (float)lr, (float)weight_decay, cur_stream());
+}
+
 torch::Tensor transpose_bf16(torch::Tensor in) {
   CHECK_BF16(in); CHECK_CONTIG(in); CHECK_CUDA(in);
   int E = in.size(0), R = in.size(1), C = in.size(2);
@@ -180,6 +209,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batched_gemm_nt", &batched_gemm_nt, "batched NT GEMM with fused epilogue",
         py::arg("A"), py::arg("B"), py::arg("bias") = py::none(),
         py::arg("pre") = py::none(), py::arg("epi") = 0);
+  m.def("batched_gemm_tn", &batched_gemm_tn, py::arg("A"), py::arg("B"), py::arg("C"),
+        py::arg("accumulate") = false);
   m.def("gating_topk_fwd", &gating_topk_fwd);
   m.def("gating_topk_bwd", &gating_topk_bwd, py::arg("probs"), py::arg("topk_idx"),
         py::arg("topk_w"), py::arg("d_topk_w"), py::arg("aux_g") = py::none());
@@ -191,6 +222,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step", &sgd_step, py::arg("master"), py::arg("grad"),
         py::arg("momentum"), py::arg("out_bf16"),
         py::arg("lr") = 0.01, py::arg("mu") = 0.0, py::arg("weight_decay") = 0.0);
+  m.def("sgd_step_t", &sgd_step_t, py::arg("master"), py::arg("grad"), py::arg("out"),
+        py::arg("outT"), py::arg("lr") = 0.01, py::arg("weight_decay") = 0.0);
   m.def("transpose_bf16", &transpose_bf16);
   m.def("colsum", &colsum);
 }

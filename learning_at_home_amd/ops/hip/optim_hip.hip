@@ -37,6 +37,53 @@ __global__ void sgd_step_kernel(float* __restrict__ master, const float* __restr
   }
 }
 
+// Fused SGD for [E, R, C] matrix params that also writes the TRANSPOSED bf16
+// copy in the same pass (saves the transpose kernel's full re-read of the
+// bf16 tensor + a launch). Plain SGD + weight decay (momentum goes through
+// the 1-D kernel + separate transpose).
+__global__ void sgd_step_t_kernel(float* __restrict__ master, const float* __restrict__ grad,
+                                  __bf16* __restrict__ out, __bf16* __restrict__ outT,
+                                  int R, int C, float lr, float weight_decay) {
+  __shared__ __bf16 tile[64][64 + 8];
+  int e = blockIdx.z;
+  long base = (long)e * R * C;
+  int r0 = blockIdx.y * 64, c0 = blockIdx.x * 64;
+  int tr = threadIdx.x >> 3;
+  int tc = (threadIdx.x & 7) * 8;
+#pragma unroll
+  for (int s = 0; s < 64; s += 32) {
+    int r = tr + s;
+    long off = base + (long)(r0 + r) * C + c0 + tc;
+    f4 m0 = *(const f4*)(master + off);
+    f4 m1 = *(const f4*)(master + off + 4);
+    f4 g0 = *(const f4*)(grad + off);
+    f4 g1 = *(const f4*)(grad + off + 4);
+    bf16x8 o;
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      m0[q] -= lr * (g0[q] + weight_decay * m0[q]);
+      m1[q] -= lr * (g1[q] + weight_decay * m1[q]);
+      o[q] = f2bf(m0[q]);
+      o[q + 4] = f2bf(m1[q]);
+    }
+    *(f4*)(master + off) = m0;
+    *(f4*)(master + off + 4) = m1;
+    *(bf16x8*)(out + off) = o;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) tile[tc + q][r] = o[q];
+  }
+  __syncthreads();
+  long baseT = (long)e * R * C;
+#pragma unroll
+  for (int s = 0; s < 64; s += 32) {
+    int r = tr + s;  // row of the transposed tile (original column)
+    bf16x8 v;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) v[q] = tile[r][tc + q];
+    *(bf16x8*)(outT + baseT + (long)(c0 + r) * R + r0 + tc) = v;
+  }
+}
+
 // batched transpose: out[e][c][r] = in[e][r][c], bf16, 64x64 LDS tiles
 // (+8-element row pad so read-back columns hit distinct banks)
 #define TP 64
@@ -88,6 +135,15 @@ extern "C" void lah_sgd_step(void* master, const void* grad, void* momentum,
   hipLaunchKernelGGL(sgd_step_kernel, dim3((unsigned)blocks), dim3(256), 0, stream,
                      (float*)master, (const float*)grad, (float*)momentum,
                      (__bf16*)out_bf16, n, lr, mu, weight_decay);
+}
+
+extern "C" void lah_sgd_step_t(void* master, const void* grad, void* out, void* outT,
+                               int E, int R, int C, float lr, float weight_decay,
+                               hipStream_t stream) {
+  dim3 grid(C / 64, R / 64, E), block(256);
+  hipLaunchKernelGGL(sgd_step_t_kernel, grid, block, 0, stream, (float*)master,
+                     (const float*)grad, (__bf16*)out, (__bf16*)outT, R, C, lr,
+                     weight_decay);
 }
 
 extern "C" void lah_transpose_bf16(const void* in, void* out, int E, int R, int C,

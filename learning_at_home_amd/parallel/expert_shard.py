@@ -65,8 +65,9 @@ class ExpertStack(nn.Module):
         self._gpu_state_ready = True
 
     def zero_main_grads(self) -> None:
-        for g in (self.main_grad_W1, self.main_grad_W2, self.main_grad_b1, self.main_grad_b2):
-            g.zero_()
+        # the TN grad kernels WRITE (not accumulate) on the first backward of
+        # a step (functional._BatchedFFNHip), so no fp32 memset is needed —
+        # clearing the flag re-arms write-mode
         self.grad_ready = False
 
 
@@ -92,25 +93,31 @@ class FusedExpertSGD:
         st = self.stack
         if not st.grad_ready:
             return
+        if self.momentum == 0.0:
+            # fused path: one pass updates master + bf16 copy + bf16^T copy
+            _C.sgd_step_t(st.W1.data, st.main_grad_W1, st.W1c, st.W1t,
+                          self.lr, self.weight_decay)
+            _C.sgd_step_t(st.W2.data, st.main_grad_W2, st.W2c, st.W2t,
+                          self.lr, self.weight_decay)
+        else:
+            mom = self.momentum_bufs
+            _C.sgd_step(st.W1.data.view(-1), st.main_grad_W1.view(-1), mom[0].view(-1),
+                        st.W1c.view(-1), self.lr, self.momentum, self.weight_decay)
+            _C.sgd_step(st.W2.data.view(-1), st.main_grad_W2.view(-1), mom[1].view(-1),
+                        st.W2c.view(-1), self.lr, self.momentum, self.weight_decay)
+            st.W1t.copy_(_C.transpose_bf16(st.W1c))
+            st.W2t.copy_(_C.transpose_bf16(st.W2c))
         mom = self.momentum_bufs or [None] * 4
-        _C.sgd_step(st.W1.data.view(-1), st.main_grad_W1.view(-1),
-                    mom[0].view(-1) if mom[0] is not None else None,
-                    st.W1c.view(-1), self.lr, self.momentum, self.weight_decay)
-        _C.sgd_step(st.W2.data.view(-1), st.main_grad_W2.view(-1),
-                    mom[1].view(-1) if mom[1] is not None else None,
-                    st.W2c.view(-1), self.lr, self.momentum, self.weight_decay)
+        scratch = torch.empty_like(st.W1_bias, dtype=torch.bfloat16)
         _C.sgd_step(st.W1_bias.data.view(-1), st.main_grad_b1.view(-1),
                     mom[2].view(-1) if mom[2] is not None else None,
-                    torch.empty_like(st.W1_bias, dtype=torch.bfloat16).view(-1),
-                    self.lr, self.momentum, self.weight_decay)
+                    scratch.view(-1), self.lr, self.momentum, self.weight_decay)
+        scratch2 = torch.empty_like(st.W2_bias, dtype=torch.bfloat16)
         _C.sgd_step(st.W2_bias.data.view(-1), st.main_grad_b2.view(-1),
                     mom[3].view(-1) if mom[3] is not None else None,
-                    torch.empty_like(st.W2_bias, dtype=torch.bfloat16).view(-1),
-                    self.lr, self.momentum, self.weight_decay)
+                    scratch2.view(-1), self.lr, self.momentum, self.weight_decay)
         st.b1_f32.copy_(st.W1_bias.data)
         st.b2_f32.copy_(st.W2_bias.data)
-        st.W1t = _C.transpose_bf16(st.W1c)
-        st.W2t = _C.transpose_bf16(st.W2c)
         st.zero_main_grads()
         self.step_count += 1
 

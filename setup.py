@@ -16,6 +16,7 @@ ext = CUDAExtension(
     sources=[
         os.path.join(HIP_DIR, "bindings.cpp"),
         os.path.join(HIP_DIR, "ffn_gemm.hip"),
+        os.path.join(HIP_DIR, "ffn_gemm_tn.hip"),
         os.path.join(HIP_DIR, "gating.hip"),
         os.path.join(HIP_DIR, "dispatch.hip"),
         os.path.join(HIP_DIR, "optim.hip"),

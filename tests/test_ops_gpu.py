@@ -70,6 +70,20 @@ class TestBatchedGemmNT:
         assert_close_bf16(C, ref, what="gelu-grad epi")
 
 
+class TestBatchedGemmTN:
+    @pytest.mark.parametrize("E,M,I,J", [(2, 64, 128, 128), (3, 192, 512, 256)])
+    def test_plain_and_accumulate(self, E, M, I, J):
+        torch.manual_seed(20)
+        A = torch.randn(E, M, I, device=DEV).bfloat16()
+        B = (torch.randn(E, M, J, device=DEV) + torch.arange(J, device=DEV) * 0.01).bfloat16()
+        C = torch.full((E, I, J), 7.0, device=DEV)  # garbage: write mode must overwrite
+        _C.batched_gemm_tn(A, B, C, False)
+        ref = torch.einsum("emi,emj->eij", A.float(), B.float())
+        assert_close_bf16(C, ref, what="gemm_tn write")
+        _C.batched_gemm_tn(A, B, C, True)
+        assert_close_bf16(C, 2 * ref, what="gemm_tn accumulate")
+
+
 class TestGating:
     @pytest.mark.parametrize("E,k", [(64, 4), (256, 4), (1024, 1), (256, 8)])
     def test_fwd(self, E, k):
@@ -209,6 +223,19 @@ class TestOptim:
         _C.sgd_step(master, grad, mom, out, 0.1, 0.9, 0.0)
         # v1=1, m1=-0.1; v2=1.9, m2=-0.29
         assert torch.allclose(master, torch.full((n,), -0.29, device=DEV), atol=1e-6)
+
+    def test_sgd_step_t_fused_transpose(self):
+        torch.manual_seed(21)
+        E, R, C = 2, 128, 64
+        master = torch.randn(E, R, C, device=DEV)
+        grad = torch.randn(E, R, C, device=DEV)
+        out = torch.empty(E, R, C, dtype=torch.bfloat16, device=DEV)
+        outT = torch.empty(E, C, R, dtype=torch.bfloat16, device=DEV)
+        ref = master - 0.05 * (grad + 0.01 * master)
+        _C.sgd_step_t(master, grad, out, outT, 0.05, 0.01)
+        assert torch.allclose(master, ref, atol=1e-6)
+        assert torch.equal(out.cpu(), ref.bfloat16().cpu())
+        assert torch.equal(outT.cpu(), ref.bfloat16().transpose(1, 2).contiguous().cpu())
 
     def test_transpose(self):
         torch.manual_seed(9)
