@@ -29,7 +29,16 @@
 #define EPI_GELU_GRAD 3   // C = acc * gelu'(PRE)
 
 // stage one [rows x BK] bf16 tile (A: rows from m0 with clamp to M; B: rows
-// from n0, N must be a multiple of BN) into LDS, lane-linear image [row][BK]
+// from n0, N must be a multiple of BN) into LDS, lane-linear image [row][BK].
+//
+// LDS XOR swizzle (guide T2 + rule 21): ds_read_b128 of column slices from a
+// row-major [128][64] bf16 image is 8-way bank-conflicted (row stride 128 B
+// ≡ 0 mod the 256-B bank row). global_load_lds writes lane-linear, so the
+// swizzle lives on the per-lane SOURCE chunk index (c8 ^ (row & 7)) with the
+// matching XOR applied at fragment-read time (lds_chunk below) — the two
+// permutations are the same involution.
+#define LDS_CHUNK(row, c) (((row) * 8 + ((c) ^ ((row) & 7))) * 8)
+
 template <bool CLAMP>
 LAH_DEVINL void stage_tile(const __bf16* __restrict__ src, __bf16* lds, int row0,
                            int max_row, int ld, int k0, int tid) {
@@ -37,7 +46,7 @@ LAH_DEVINL void stage_tile(const __bf16* __restrict__ src, __bf16* lds, int row0
   for (int i = 0; i < (BM * BK) / (NTHREADS * 8); ++i) {
     int idx = i * NTHREADS + tid;       // 16-byte chunk index
     int row = idx >> 3;                 // BK=64 -> 8 chunks per row
-    int c8 = idx & 7;
+    int c8 = (idx & 7) ^ (row & 7);     // inverse swizzle on the source side
     int grow = row0 + row;
     if (CLAMP) grow = min(grow, max_row - 1);
     glds16(src + (long)grow * ld + k0 + c8 * 8, lds + idx * 8);
@@ -92,12 +101,17 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
 #pragma unroll
     for (int kk = 0; kk < BK / 32; ++kk) {
       bf16x8 afrag[4], bfrag[4];
+      const int kchunk = (kk * 32 + lk8) >> 3;
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        afrag[mi] = *(const bf16x8*)(la + (wr * 64 + mi * 16 + l15) * BK + kk * 32 + lk8);
+      for (int mi = 0; mi < 4; ++mi) {
+        int row = wr * 64 + mi * 16 + l15;
+        afrag[mi] = *(const bf16x8*)(la + LDS_CHUNK(row, kchunk));
+      }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
-        bfrag[ni] = *(const bf16x8*)(lb + (wc * 64 + ni * 16 + l15) * BK + kk * 32 + lk8);
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = wc * 64 + ni * 16 + l15;
+        bfrag[ni] = *(const bf16x8*)(lb + LDS_CHUNK(row, kchunk));
+      }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll

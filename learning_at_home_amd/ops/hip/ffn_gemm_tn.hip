@@ -16,7 +16,13 @@
 #define TBK 64  // reduction (m) tile
 #define TN_THREADS 256
 
-// stage one [TBK rows of m][128 cols] tile TRANSPOSED into LDS image [128][TBK]
+// stage one [TBK rows of m][128 cols] tile TRANSPOSED into LDS image
+// [128][TBK]. XOR swizzle on the m-chunk index spreads both the scalar
+// transpose writes (which vary only in i>>3 within an instruction) and the
+// b128 fragment reads (which vary in i) over distinct banks (guide T2/G4):
+//   chunk' = (m>>3) ^ (i & 7) ^ ((i >> 3) & 7)
+#define TN_OFF(i, m) ((i) * TBK + (((((m) >> 3) ^ ((i) & 7) ^ (((i) >> 3) & 7)) << 3) | ((m) & 7)))
+
 LAH_DEVINL void stage_tile_t(const __bf16* __restrict__ src, __bf16* lds, int col0,
                              int m0, int ld, int tid) {
 #pragma unroll
@@ -26,7 +32,7 @@ LAH_DEVINL void stage_tile_t(const __bf16* __restrict__ src, __bf16* lds, int co
     int i16 = idx & 15;
     bf16x8 v = *(const bf16x8*)(src + (long)(m0 + m) * ld + col0 + i16 * 8);
 #pragma unroll
-    for (int q = 0; q < 8; ++q) lds[(i16 * 8 + q) * TBK + m] = v[q];
+    for (int q = 0; q < 8; ++q) lds[TN_OFF(i16 * 8 + q, m)] = v[q];
   }
 }
 
@@ -75,12 +81,17 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
 #pragma unroll
     for (int kk = 0; kk < TBK / 32; ++kk) {
       bf16x8 afrag[4], bfrag[4];
+      const int km = kk * 32 + lk8;
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
-        afrag[mi] = *(const bf16x8*)(la + (wr * 64 + mi * 16 + l15) * TBK + kk * 32 + lk8);
+      for (int mi = 0; mi < 4; ++mi) {
+        int row = wr * 64 + mi * 16 + l15;
+        afrag[mi] = *(const bf16x8*)(la + TN_OFF(row, km));
+      }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni)
-        bfrag[ni] = *(const bf16x8*)(lb + (wc * 64 + ni * 16 + l15) * TBK + kk * 32 + lk8);
+      for (int ni = 0; ni < 4; ++ni) {
+        int row = wc * 64 + ni * 16 + l15;
+        bfrag[ni] = *(const bf16x8*)(lb + TN_OFF(row, km));
+      }
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi)
 #pragma unroll
