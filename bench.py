@@ -36,10 +36,9 @@ def build_model(cfg: TransformerDMoEConfig, args, device, group=None):
     model = TransformerDMoE(cfg, moe_factory)
     if device.type == "cuda":
         model = model.to(device=device, dtype=torch.bfloat16)
-        # fp32 masters for expert stacks + LN in fp32 for stability
         for block in model.blocks:
             block.moe.experts.float()
-            block.moe.experts.init_gpu_state()
+            block.moe.experts.init_gpu_state(master_weights=args.master_weights)
     else:
         model = model.to(device)
     return model
@@ -64,6 +63,9 @@ def main():
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--averaging-period", type=int, default=10)
     p.add_argument("--no-averager", action="store_true")
+    p.add_argument("--master-weights", action="store_true",
+                   help="fp32 master weights for experts (default: bf16 + "
+                        "stochastic-rounding SGD, the fast path)")
     args = p.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -103,7 +105,8 @@ def main():
             trunk_params.append(param)
     for block in model.blocks:
         if device.type == "cuda" and HAS_HIP:
-            expert_opts.append(FusedExpertSGD(block.moe.experts, lr=args.lr))
+            expert_opts.append(FusedExpertSGD(block.moe.experts, lr=args.lr,
+                                              stochastic_rounding=not args.master_weights))
         else:
             expert_opts.append(torch.optim.SGD(block.moe.experts.parameters(), lr=args.lr))
     trunk_opt = torch.optim.SGD(trunk_params, lr=args.lr)

@@ -199,20 +199,21 @@ class _BatchedFFNHip(torch.autograd.Function):
     @staticmethod
     def forward(ctx, xbuf: torch.Tensor, stack):
         E, C, D = xbuf.shape
-        pre1 = torch.empty(E, C, stack.ffn_dim, dtype=torch.bfloat16, device=xbuf.device)
-        h = _C.batched_gemm_nt(xbuf.contiguous(), stack.W1c, stack.b1_f32, pre1, EPI_BIAS_GELU)
+        # der = gelu'(pre-act), computed for free in the fwd epilogue (one tanh)
+        der = torch.empty(E, C, stack.ffn_dim, dtype=torch.bfloat16, device=xbuf.device)
+        h = _C.batched_gemm_nt(xbuf.contiguous(), stack.W1c, stack.b1_f32, der, EPI_BIAS_GELU)
         y = _C.batched_gemm_nt(h, stack.W2c, stack.b2_f32, None, EPI_BIAS)
-        ctx.save_for_backward(xbuf, pre1, h)
+        ctx.save_for_backward(xbuf, der, h)
         ctx.stack = stack
         return y
 
     @staticmethod
     def backward(ctx, dy: torch.Tensor):
-        xbuf, pre1, h = ctx.saved_tensors
+        xbuf, der, h = ctx.saved_tensors
         stack = ctx.stack
         dy = dy.contiguous()
-        # dh_pre = (dy @ W2t^T) * gelu'(pre1); dx = dh_pre @ W1t^T   (MFMA NT)
-        dh_pre = _C.batched_gemm_nt(dy, stack.W2t, None, pre1, EPI_GELU_GRAD)
+        # dh_pre = (dy @ W2t^T) * der; dx = dh_pre @ W1t^T   (MFMA NT)
+        dh_pre = _C.batched_gemm_nt(dy, stack.W2t, None, der, EPI_GELU_GRAD)
         dxbuf = _C.batched_gemm_nt(dh_pre, stack.W1t, None, None, EPI_NONE)
         # weight grads: hand-written TN MFMA kernels write the fp32 grad
         # buffers in place (accumulate only after the first backward of a step)

@@ -26,6 +26,8 @@ void lah_sgd_step(void*, const void*, void*, void*, long, float, float, float,
                   hipStream_t);
 void lah_sgd_step_t(void*, const void*, void*, void*, int, int, int, float, float,
                     hipStream_t);
+void lah_sgd_step_sr_t(void*, const void*, void*, int, int, int, float, float,
+                       unsigned, hipStream_t);
 void lah_transpose_bf16(const void*, void*, int, int, int, hipStream_t);
 void lah_colsum(const void*, void*, int, int, int, hipStream_t);
 }
@@ -186,6 +188,16 @@ void sgd_step_t(torch::Tensor master, torch::Tensor grad, torch::Tensor out,
                  E, R, C, (float)lr, (float)weight_decay, cur_stream());
 }
 
+void sgd_step_sr_t(torch::Tensor param, torch::Tensor grad, torch::Tensor paramT,
+                   double lr, double weight_decay, int64_t seed) {
+  CHECK_BF16(param); CHECK_F32(grad); CHECK_BF16(paramT);
+  CHECK_CONTIG(param); CHECK_CONTIG(grad); CHECK_CONTIG(paramT);
+  int E = param.size(0), R = param.size(1), C = param.size(2);
+  TORCH_CHECK(R % 64 == 0 && C % 64 == 0, "dims must be multiples of 64");
+  lah_sgd_step_sr_t(param.data_ptr(), grad.data_ptr(), paramT.data_ptr(), E, R, C,
+                    (float)lr, (float)weight_decay, (unsigned)seed, cur_stream());
+}
+
 torch::Tensor transpose_bf16(torch::Tensor in) {
   CHECK_BF16(in); CHECK_CONTIG(in); CHECK_CUDA(in);
   int E = in.size(0), R = in.size(1), C = in.size(2);
@@ -224,6 +236,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("lr") = 0.01, py::arg("mu") = 0.0, py::arg("weight_decay") = 0.0);
   m.def("sgd_step_t", &sgd_step_t, py::arg("master"), py::arg("grad"), py::arg("out"),
         py::arg("outT"), py::arg("lr") = 0.01, py::arg("weight_decay") = 0.0);
+  m.def("sgd_step_sr_t", &sgd_step_sr_t, py::arg("param"), py::arg("grad"),
+        py::arg("paramT"), py::arg("lr") = 0.01, py::arg("weight_decay") = 0.0,
+        py::arg("seed") = 0);
   m.def("transpose_bf16", &transpose_bf16);
   m.def("colsum", &colsum);
 }

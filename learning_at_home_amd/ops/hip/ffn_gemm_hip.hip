@@ -4,47 +4,38 @@
 //
 // One kernel family: batched NT GEMM C[e] = A[e] @ B[e]^T with fused
 // epilogues, covering all four activation-side GEMMs of the expert FFN:
-//   fwd1: h   = gelu(x @ W1^T + b1)          (EPI_BIAS_GELU, stores pre-act)
-//   fwd2: y   = h @ W2^T + b2                (EPI_BIAS)
-//   bwd1: dhp = (dy @ W2t^T) * gelu'(pre)    (EPI_GELU_GRAD, reads pre-act)
-//   bwd2: dx  = dhp @ W1t^T                  (EPI_NONE)
-// (W1t/W2t are transposed bf16 weight copies maintained by the fused
-// optimizer so every GEMM is the MFMA-natural NT layout — both operands
-// K-major; see optim.hip.)
+//   fwd1: h   = gelu(x @ W1^T + b1)        (EPI_BIAS_GELU, stores gelu'(pre))
+//   fwd2: y   = h @ W2^T + b2              (EPI_BIAS)
+//   bwd1: dhp = (dy @ W2t^T) * DER         (EPI_GELU_GRAD, reads gelu'(pre))
+//   bwd2: dx  = dhp @ W1t^T                (EPI_NONE)
+// The fwd GELU epilogue stores the DERIVATIVE gelu'(pre) (not the pre-act):
+// it reuses the tanh it already computed, making the backward epilogue a
+// pure multiply — no transcendentals in bwd.
 //
-// Structure = the "step-3" ladder structure of the CDNA4 guide (§5):
-// 128x128 tile, BK=64, 4 waves of 64x64, mfma_f32_16x16x32_bf16,
-// double-buffered LDS staged by global_load_lds (16 B/lane), prefetch
-// next K-tile before computing current. Weight-tile (B) rows are L2-reused
-// across the M-tiles of the same expert.
+// Structure: guide §5 "step-3" + T2 swizzle: LDS-staged via global_load_lds
+// (16 B/lane), double-buffered, mfma_f32_16x16x32_bf16, XOR chunk swizzle on
+// the glds SOURCE address with the matching XOR at fragment read (rule 21).
+// Tile geometry is templated: 128x128 (4 waves of 64x64) for M%128==0,
+// 64x128 (4 waves of 64x32) otherwise — capacity buffers are padded to 64
+// rows, so a 192-row expert batch runs as 3 full 64-tiles with zero waste.
 #include "common.h"
 
-#define BM 128
-#define BN 128
 #define BK 64
 #define NTHREADS 256
 
 #define EPI_NONE 0
 #define EPI_BIAS 1
-#define EPI_BIAS_GELU 2   // C = gelu(acc + bias), PRE = acc + bias
-#define EPI_GELU_GRAD 3   // C = acc * gelu'(PRE)
+#define EPI_BIAS_GELU 2   // C = gelu(acc + bias), DER = gelu'(acc + bias)
+#define EPI_GELU_GRAD 3   // C = acc * DER
 
-// stage one [rows x BK] bf16 tile (A: rows from m0 with clamp to M; B: rows
-// from n0, N must be a multiple of BN) into LDS, lane-linear image [row][BK].
-//
-// LDS XOR swizzle (guide T2 + rule 21): ds_read_b128 of column slices from a
-// row-major [128][64] bf16 image is 8-way bank-conflicted (row stride 128 B
-// ≡ 0 mod the 256-B bank row). global_load_lds writes lane-linear, so the
-// swizzle lives on the per-lane SOURCE chunk index (c8 ^ (row & 7)) with the
-// matching XOR applied at fragment-read time (lds_chunk below) — the two
-// permutations are the same involution.
+// swizzled LDS element offset of 16-byte chunk c in row `row` (guide T2)
 #define LDS_CHUNK(row, c) (((row) * 8 + ((c) ^ ((row) & 7))) * 8)
 
-template <bool CLAMP>
+template <int ROWS, bool CLAMP>
 LAH_DEVINL void stage_tile(const __bf16* __restrict__ src, __bf16* lds, int row0,
                            int max_row, int ld, int k0, int tid) {
 #pragma unroll
-  for (int i = 0; i < (BM * BK) / (NTHREADS * 8); ++i) {
+  for (int i = 0; i < (ROWS * BK) / (NTHREADS * 8); ++i) {
     int idx = i * NTHREADS + tid;       // 16-byte chunk index
     int row = idx >> 3;                 // BK=64 -> 8 chunks per row
     int c8 = (idx & 7) ^ (row & 7);     // inverse swizzle on the source side
@@ -54,14 +45,29 @@ LAH_DEVINL void stage_tile(const __bf16* __restrict__ src, __bf16* lds, int row0
   }
 }
 
-template <int EPI>
+// gelu and its derivative from one tanh evaluation
+LAH_DEVINL void gelu_both(float x, float* g, float* dg) {
+  float x2 = x * x;
+  float u = GELU_C0 * (x + GELU_C1 * x * x2);
+  float t = tanhf(u);
+  float du = GELU_C0 * (1.0f + 3.0f * GELU_C1 * x2);
+  *g = 0.5f * x * (1.0f + t);
+  *dg = 0.5f * (1.0f + t) + 0.5f * x * (1.0f - t * t) * du;
+}
+
+// WM x WN waves, each computing FM x FN fragments of 16x16:
+// BM = WM*FM*16, BN = WN*FN*16. Instantiated as <EPI, 2,2,4,4> (128x128)
+// and <EPI, 1,4,4,2> (64x128).
+template <int EPI, int WM, int WN, int FM, int FN>
 __launch_bounds__(NTHREADS, 2) __global__
 void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
                             const __bf16* __restrict__ B,  // [E, N, K]
                             __bf16* __restrict__ C,        // [E, M, N]
                             const float* __restrict__ bias,  // [E, N] or null
-                            __bf16* __restrict__ PRE,      // [E, M, N] store/read
+                            __bf16* __restrict__ DER,      // [E, M, N] store/read
                             int M, int N, int K) {
+  constexpr int BM = WM * FM * 16;
+  constexpr int BN = WN * FN * 16;
   __shared__ __bf16 smem[2 * (BM + BN) * BK];
   const int HALF = (BM + BN) * BK;
 
@@ -71,21 +77,21 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wr = wave >> 1, wc = wave & 1;  // wave's 64x64 sub-tile
+  const int wr = wave / WN, wc = wave % WN;
 
   const __bf16* Ae = A + (long)e * M * K;
   const __bf16* Be = B + (long)e * N * K;
 
-  f32x4 acc[4][4];
+  f32x4 acc[FM][FN];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < FM; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int nk = K / BK;
   int cur = 0;
-  stage_tile<true>(Ae, smem, m0, M, K, 0, tid);
-  stage_tile<false>(Be, smem + BM * BK, n0, N, K, 0, tid);
+  stage_tile<BM, true>(Ae, smem, m0, M, K, 0, tid);
+  stage_tile<BN, false>(Be, smem + BM * BK, n0, N, K, 0, tid);
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
 
@@ -94,29 +100,30 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
 
   for (int t = 0; t < nk; ++t) {
     if (t + 1 < nk) {
-      stage_tile<true>(Ae, smem + (cur ^ 1) * HALF, m0, M, K, (t + 1) * BK, tid);
-      stage_tile<false>(Be, smem + (cur ^ 1) * HALF + BM * BK, n0, N, K, (t + 1) * BK, tid);
+      stage_tile<BM, true>(Ae, smem + (cur ^ 1) * HALF, m0, M, K, (t + 1) * BK, tid);
+      stage_tile<BN, false>(Be, smem + (cur ^ 1) * HALF + BM * BK, n0, N, K,
+                            (t + 1) * BK, tid);
     }
     const __bf16* la = smem + cur * HALF;
     const __bf16* lb = smem + cur * HALF + BM * BK;
 #pragma unroll
     for (int kk = 0; kk < BK / 32; ++kk) {
-      bf16x8 afrag[4], bfrag[4];
+      bf16x8 afrag[FM], bfrag[FN];
       const int kchunk = (kk * 32 + lk8) >> 3;
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
-        int row = wr * 64 + mi * 16 + l15;
+      for (int mi = 0; mi < FM; ++mi) {
+        int row = wr * FM * 16 + mi * 16 + l15;
         afrag[mi] = *(const bf16x8*)(la + LDS_CHUNK(row, kchunk));
       }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
-        int row = wc * 64 + ni * 16 + l15;
+      for (int ni = 0; ni < FN; ++ni) {
+        int row = wc * FN * 16 + ni * 16 + l15;
         bfrag[ni] = *(const bf16x8*)(lb + LDS_CHUNK(row, kchunk));
       }
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < FM; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
+        for (int ni = 0; ni < FN; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[mi], bfrag[ni],
                                                                 acc[mi][ni], 0, 0, 0);
     }
@@ -128,28 +135,29 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
   // epilogue: C/D fragment mapping for 16x16x32: col = lane&15,
   // row = (lane>>4)*4 + reg  (cdna_hip_programming.md §3)
   __bf16* Ce = C + (long)e * M * N;
-  __bf16* Pe = (EPI == EPI_BIAS_GELU || EPI == EPI_GELU_GRAD)
-                   ? PRE + (long)e * M * N : nullptr;
+  __bf16* De = (EPI == EPI_BIAS_GELU || EPI == EPI_GELU_GRAD)
+                   ? DER + (long)e * M * N : nullptr;
   const float* be = (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) ? bias + (long)e * N : nullptr;
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
+  for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-      int col = n0 + wc * 64 + ni * 16 + l15;
+    for (int ni = 0; ni < FN; ++ni) {
+      int col = n0 + wc * FN * 16 + ni * 16 + l15;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int row = m0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+        int row = m0 + wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
         if (row >= M) continue;
         long off = (long)row * N + col;
         float v = acc[mi][ni][r];
         if (EPI == EPI_BIAS) {
           Ce[off] = f2bf(v + be[col]);
         } else if (EPI == EPI_BIAS_GELU) {
-          float pre = v + be[col];
-          Pe[off] = f2bf(pre);
-          Ce[off] = f2bf(gelu_fwd(pre));
+          float g, dg;
+          gelu_both(v + be[col], &g, &dg);
+          Ce[off] = f2bf(g);
+          De[off] = f2bf(dg);
         } else if (EPI == EPI_GELU_GRAD) {
-          Ce[off] = f2bf(v * gelu_bwd(bf2f(Pe[off])));
+          Ce[off] = f2bf(v * bf2f(De[off]));
         } else {
           Ce[off] = f2bf(v);
         }
@@ -158,33 +166,33 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
   }
 }
 
-extern "C" void lah_batched_gemm_nt(const void* A, const void* B, void* C,
-                                    const void* bias, void* PRE, int E, int M, int N,
-                                    int K, int epi, hipStream_t stream) {
-  dim3 grid(N / BN, cdiv(M, BM), E);
+template <int EPI>
+static void launch_nt(const void* A, const void* B, void* C, const void* bias,
+                      void* DER, int E, int M, int N, int K, hipStream_t stream) {
   dim3 block(NTHREADS);
+  if (M % 128 == 0) {
+    dim3 grid(N / 128, M / 128, E);
+    hipLaunchKernelGGL((batched_gemm_nt_kernel<EPI, 2, 2, 4, 4>), grid, block, 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
+                       (const float*)bias, (__bf16*)DER, M, N, K);
+  } else {  // M padded to 64 (capacity buffers): 64-row tiles, zero tail waste
+    dim3 grid(N / 128, cdiv(M, 64), E);
+    hipLaunchKernelGGL((batched_gemm_nt_kernel<EPI, 1, 4, 4, 2>), grid, block, 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
+                       (const float*)bias, (__bf16*)DER, M, N, K);
+  }
+}
+
+extern "C" void lah_batched_gemm_nt(const void* A, const void* B, void* C,
+                                    const void* bias, void* DER, int E, int M, int N,
+                                    int K, int epi, hipStream_t stream) {
   switch (epi) {
-    case EPI_NONE:
-      hipLaunchKernelGGL(batched_gemm_nt_kernel<EPI_NONE>, grid, block, 0, stream,
-                         (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
-                         (const float*)bias, (__bf16*)PRE, M, N, K);
-      break;
-    case EPI_BIAS:
-      hipLaunchKernelGGL(batched_gemm_nt_kernel<EPI_BIAS>, grid, block, 0, stream,
-                         (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
-                         (const float*)bias, (__bf16*)PRE, M, N, K);
-      break;
+    case EPI_NONE: launch_nt<EPI_NONE>(A, B, C, bias, DER, E, M, N, K, stream); break;
+    case EPI_BIAS: launch_nt<EPI_BIAS>(A, B, C, bias, DER, E, M, N, K, stream); break;
     case EPI_BIAS_GELU:
-      hipLaunchKernelGGL(batched_gemm_nt_kernel<EPI_BIAS_GELU>, grid, block, 0, stream,
-                         (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
-                         (const float*)bias, (__bf16*)PRE, M, N, K);
-      break;
+      launch_nt<EPI_BIAS_GELU>(A, B, C, bias, DER, E, M, N, K, stream); break;
     case EPI_GELU_GRAD:
-      hipLaunchKernelGGL(batched_gemm_nt_kernel<EPI_GELU_GRAD>, grid, block, 0, stream,
-                         (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
-                         (const float*)bias, (__bf16*)PRE, M, N, K);
-      break;
-    default:
-      abort();
+      launch_nt<EPI_GELU_GRAD>(A, B, C, bias, DER, E, M, N, K, stream); break;
+    default: abort();
   }
 }

@@ -83,6 +83,67 @@ __global__ void sgd_step_t_kernel(float* __restrict__ master, const float* __res
   }
 }
 
+// Master-free SGD: bf16 params updated in place with STOCHASTIC ROUNDING
+// (E[rounded] = exact value, so small updates are preserved in expectation
+// without an fp32 master copy — 10 B/param of traffic instead of 16) + the
+// transposed bf16 copy in the same pass.
+LAH_DEVINL unsigned hash_u32(unsigned x) {
+  x ^= x >> 16; x *= 0x7feb352du; x ^= x >> 15; x *= 0x846ca68bu; x ^= x >> 16;
+  return x;
+}
+
+LAH_DEVINL __bf16 round_sr(float v, unsigned rnd16) {
+  unsigned bits = __float_as_uint(v);
+  unsigned frac = bits & 0xFFFFu;
+  unsigned trunc = bits & 0xFFFF0000u;
+  if (frac > (rnd16 & 0xFFFFu)) trunc += 0x10000u;
+  unsigned short hi = (unsigned short)(trunc >> 16);
+  return *reinterpret_cast<__bf16*>(&hi);
+}
+
+__global__ void sgd_step_sr_t_kernel(__bf16* __restrict__ param, const float* __restrict__ grad,
+                                     __bf16* __restrict__ paramT, int R, int C, float lr,
+                                     float weight_decay, unsigned seed) {
+  __shared__ __bf16 tile[64][64 + 8];
+  int e = blockIdx.z;
+  long base = (long)e * R * C;
+  int r0 = blockIdx.y * 64, c0 = blockIdx.x * 64;
+  int tr = threadIdx.x >> 3;
+  int tc = (threadIdx.x & 7) * 8;
+#pragma unroll
+  for (int s = 0; s < 64; s += 32) {
+    int r = tr + s;
+    long off = base + (long)(r0 + r) * C + c0 + tc;
+    bf16x8 p = *(const bf16x8*)(param + off);
+    f4 g0 = *(const f4*)(grad + off);
+    f4 g1 = *(const f4*)(grad + off + 4);
+    bf16x8 o;
+    unsigned h = hash_u32((unsigned)(off & 0xFFFFFFFFu) ^ seed);
+    unsigned h2 = hash_u32(h + 0x9e3779b9u);
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      float pv = bf2f(p[q]);
+      pv -= lr * (g0[q] + weight_decay * pv);
+      o[q] = round_sr(pv, (q < 2 ? h : h2) >> (16 * (q & 1)));
+      float pv2 = bf2f(p[q + 4]);
+      pv2 -= lr * (g1[q] + weight_decay * pv2);
+      o[q + 4] = round_sr(pv2, hash_u32(h + q) >> 8);
+    }
+    *(bf16x8*)(param + off) = o;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) tile[tc + q][r] = o[q];
+  }
+  __syncthreads();
+#pragma unroll
+  for (int s = 0; s < 64; s += 32) {
+    int r = tr + s;
+    bf16x8 v;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) v[q] = tile[r][tc + q];
+    *(bf16x8*)(paramT + base + (long)(c0 + r) * R + r0 + tc) = v;
+  }
+}
+
 // batched transpose: out[e][c][r] = in[e][r][c], bf16, 64x64 LDS tiles
 // (+8-element row pad so read-back columns hit distinct banks)
 #define TP 64
@@ -143,6 +204,14 @@ extern "C" void lah_sgd_step_t(void* master, const void* grad, void* out, void* 
   hipLaunchKernelGGL(sgd_step_t_kernel, grid, block, 0, stream, (float*)master,
                      (const float*)grad, (__bf16*)out, (__bf16*)outT, R, C, lr,
                      weight_decay);
+}
+
+extern "C" void lah_sgd_step_sr_t(void* param, const void* grad, void* paramT, int E,
+                                  int R, int C, float lr, float weight_decay,
+                                  unsigned seed, hipStream_t stream) {
+  dim3 grid(C / 64, R / 64, E), block(256);
+  hipLaunchKernelGGL(sgd_step_sr_t_kernel, grid, block, 0, stream, (__bf16*)param,
+                     (const float*)grad, (__bf16*)paramT, R, C, lr, weight_decay, seed);
 }
 
 extern "C" void lah_transpose_bf16(const void* in, void* out, int E, int R, int C,

@@ -48,21 +48,36 @@ class ExpertStack(nn.Module):
         self.grad_ready = False
 
     # ---- GPU compute-state management ---------------------------------
-    def init_gpu_state(self) -> None:
-        """Allocate bf16 compute/transposed copies + fp32 grad buffers."""
+    def init_gpu_state(self, master_weights: bool = True) -> None:
+        """Allocate bf16 compute/transposed copies + fp32 grad buffers.
+        master_weights=False → stochastic-rounding mode: the fp32 masters of
+        W1/W2 move to host (updates happen directly on the bf16 copies with
+        unbiased rounding — see optim.hip sgd_step_sr_t); call
+        sync_master_weights() before checkpointing in that mode."""
         assert self.W1.is_cuda, "move the stack to GPU first"
         dev = self.W1.device
+        self.master_weights = master_weights
         self.W1c = self.W1.detach().bfloat16().contiguous()
         self.W2c = self.W2.detach().bfloat16().contiguous()
         self.W1t = _C.transpose_bf16(self.W1c)  # [E, d, ffn]
         self.W2t = _C.transpose_bf16(self.W2c)  # [E, ffn, d]
         self.b1_f32 = self.W1_bias.detach().float().contiguous()
         self.b2_f32 = self.W2_bias.detach().float().contiguous()
-        self.main_grad_W1 = torch.zeros_like(self.W1, device=dev)
-        self.main_grad_W2 = torch.zeros_like(self.W2, device=dev)
+        if not master_weights:
+            self.W1.data = self.W1.data.cpu()  # free 2x HBM for weights
+            self.W2.data = self.W2.data.cpu()
+        self.main_grad_W1 = torch.zeros(self.W1.shape, device=dev)
+        self.main_grad_W2 = torch.zeros(self.W2.shape, device=dev)
         self.main_grad_b1 = torch.zeros_like(self.W1_bias, device=dev)
         self.main_grad_b2 = torch.zeros_like(self.W2_bias, device=dev)
         self._gpu_state_ready = True
+
+    def sync_master_weights(self) -> None:
+        """Pull bf16 compute weights back into the fp32 master Parameters
+        (needed before state_dict/checkpoint in stochastic-rounding mode)."""
+        if self._gpu_state_ready and not self.master_weights:
+            self.W1.data = self.W1c.detach().float().to(self.W1.data.device)
+            self.W2.data = self.W2c.detach().float().to(self.W2.data.device)
 
     def zero_main_grads(self) -> None:
         # the TN grad kernels WRITE (not accumulate) on the first backward of
@@ -77,9 +92,14 @@ class FusedExpertSGD:
     refreshed by the tiled transpose kernel. (SURVEY.md §2 hot-op #5.)"""
 
     def __init__(self, stack: ExpertStack, lr: float = 0.01, momentum: float = 0.0,
-                 weight_decay: float = 0.0):
+                 weight_decay: float = 0.0, stochastic_rounding: bool = False):
         self.stack = stack
         self.lr, self.momentum, self.weight_decay = lr, momentum, weight_decay
+        self.stochastic_rounding = stochastic_rounding
+        if stochastic_rounding:
+            assert momentum == 0.0, "SR mode supports plain SGD only"
+            assert not getattr(stack, "master_weights", True), (
+                "init_gpu_state(master_weights=False) first for SR mode")
         self.momentum_bufs = None
         if momentum > 0:
             self.momentum_bufs = [
@@ -93,7 +113,13 @@ class FusedExpertSGD:
         st = self.stack
         if not st.grad_ready:
             return
-        if self.momentum == 0.0:
+        if self.stochastic_rounding:
+            seed = (self.step_count * 2654435761) & 0x7FFFFFFF
+            _C.sgd_step_sr_t(st.W1c, st.main_grad_W1, st.W1t, self.lr,
+                             self.weight_decay, seed)
+            _C.sgd_step_sr_t(st.W2c, st.main_grad_W2, st.W2t, self.lr,
+                             self.weight_decay, seed ^ 0x55555555)
+        elif self.momentum == 0.0:
             # fused path: one pass updates master + bf16 copy + bf16^T copy
             _C.sgd_step_t(st.W1.data, st.main_grad_W1, st.W1c, st.W1t,
                           self.lr, self.weight_decay)
