@@ -1,0 +1,141 @@
+"""DecentralizedAverager tests: DHT matchmaking + butterfly
+part all-reduce over sockets on localhost; RcclAverager via gloo (CPU)."""
+import os
+import threading
+import time
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from learning_at_home_amd import DHT
+from learning_at_home_amd.averaging import DecentralizedAverager, RcclAverager
+from learning_at_home_amd.optim import DecentralizedSGD
+
+
+class TestDecentralizedAverager:
+    def test_two_peer_round(self):
+        dht = DHT(start=True)
+        dht2 = DHT(initial_peers=[dht.endpoint], start=True)
+        t1 = [torch.ones(16), torch.full((4, 4), 2.0)]
+        t2 = [torch.full((16,), 3.0), torch.full((4, 4), 6.0)]
+        a1 = DecentralizedAverager(t1, dht, prefix="avgtest", target_group_size=2,
+                                   averaging_expiration=1.0, start=True)
+        a2 = DecentralizedAverager(t2, dht2, prefix="avgtest", target_group_size=2,
+                                   averaging_expiration=1.0, start=True)
+        results = {}
+
+        def run(name, av):
+            results[name] = av.step(timeout=20.0)
+
+        th1 = threading.Thread(target=run, args=("a", a1))
+        th2 = threading.Thread(target=run, args=("b", a2))
+        th1.start(); time.sleep(0.2); th2.start()
+        th1.join(25); th2.join(25)
+        try:
+            assert results.get("a") is not None and results.get("b") is not None, (
+                f"matchmaking failed: {results}")
+            for tensors in (t1, t2):
+                assert torch.allclose(tensors[0], torch.full((16,), 2.0), atol=1e-5)
+                assert torch.allclose(tensors[1], torch.full((4, 4), 4.0), atol=1e-5)
+        finally:
+            a1.shutdown(); a2.shutdown(); dht2.shutdown(); dht.shutdown()
+
+    def test_weighted_average(self):
+        dht = DHT(start=True)
+        dht2 = DHT(initial_peers=[dht.endpoint], start=True)
+        t1 = [torch.zeros(8)]
+        t2 = [torch.ones(8)]
+        a1 = DecentralizedAverager(t1, dht, prefix="wavg", target_group_size=2,
+                                   averaging_expiration=1.0, start=True)
+        a2 = DecentralizedAverager(t2, dht2, prefix="wavg", target_group_size=2,
+                                   averaging_expiration=1.0, start=True)
+        results = {}
+
+        def run(name, av, w):
+            results[name] = av.step(weight=w, timeout=20.0)
+
+        th1 = threading.Thread(target=run, args=("a", a1, 3.0))
+        th2 = threading.Thread(target=run, args=("b", a2, 1.0))
+        th1.start(); time.sleep(0.2); th2.start()
+        th1.join(25); th2.join(25)
+        try:
+            assert results.get("a") is not None and results.get("b") is not None
+            # weighted: (3*0 + 1*1) / 4 = 0.25
+            assert torch.allclose(t1[0], torch.full((8,), 0.25), atol=1e-5)
+            assert torch.allclose(t2[0], torch.full((8,), 0.25), atol=1e-5)
+        finally:
+            a1.shutdown(); a2.shutdown(); dht2.shutdown(); dht.shutdown()
+
+    def test_solo_step_returns_none(self):
+        dht = DHT(start=True)
+        t = [torch.ones(4)]
+        a = DecentralizedAverager(t, dht, prefix="solo", target_group_size=2,
+                                  averaging_expiration=0.5, start=True)
+        try:
+            assert a.step(timeout=3.0) is None
+            assert torch.equal(t[0], torch.ones(4))  # unchanged
+        finally:
+            a.shutdown(); dht.shutdown()
+
+    def test_load_state_from_peers(self):
+        dht = DHT(start=True)
+        dht2 = DHT(initial_peers=[dht.endpoint], start=True)
+        from learning_at_home_amd.utils import get_dht_time
+
+        src = DecentralizedAverager([torch.full((6,), 42.0)], dht, prefix="state",
+                                    averaging_expiration=5.0, start=True)
+        # src advertises under the group leader key so the newcomer finds it
+        dht.store(src._leader_key, list(src.endpoint), get_dht_time() + 30)
+        dst = DecentralizedAverager([torch.zeros(6)], dht2, prefix="state",
+                                    averaging_expiration=5.0, start=True)
+        try:
+            ok = dst.load_state_from_peers(timeout=10.0)
+            assert ok, "state transfer failed"
+            with dst.get_tensors() as tensors:
+                assert torch.equal(tensors[0], torch.full((6,), 42.0))
+        finally:
+            src.shutdown(); dst.shutdown(); dht2.shutdown(); dht.shutdown()
+
+
+def _rccl_worker(rank, world):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29531")
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    params = [torch.full((10,), float(rank)), torch.full((3, 3), float(rank * 10))]
+    av = RcclAverager(params, dtype=torch.float32)
+    av.step(async_op=False)
+    expected0 = sum(range(world)) / world
+    expected1 = sum(10.0 * r for r in range(world)) / world
+    assert torch.allclose(params[0], torch.full((10,), expected0)), params[0]
+    assert torch.allclose(params[1], torch.full((3, 3), expected1))
+    dist.destroy_process_group()
+
+
+def _dsgd_worker(rank, world):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29532")
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(rank)
+    model = torch.nn.Linear(4, 4)
+    av = RcclAverager(list(model.parameters()), dtype=torch.float32)
+    opt = DecentralizedSGD(model.parameters(), lr=0.05, averager=av, averaging_period=2)
+    for _ in range(4):
+        loss = model(torch.randn(8, 4)).pow(2).mean()
+        opt.zero_grad(); loss.backward(); opt.step()
+    av.wait()
+    dist.barrier()
+    # after an averaging round, replicas should agree
+    flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+    gathered = [torch.zeros_like(flat) for _ in range(world)]
+    dist.all_gather(gathered, flat)
+    for g in gathered[1:]:
+        assert torch.allclose(gathered[0], g, atol=1e-5)
+    dist.destroy_process_group()
+
+
+class TestRcclAverager:
+    def test_gloo_two_ranks(self):
+        mp.spawn(_rccl_worker, args=(2,), nprocs=2, join=True)
+
+    def test_decentralized_sgd_converges_replicas(self):
+        mp.spawn(_dsgd_worker, args=(2,), nprocs=2, join=True)

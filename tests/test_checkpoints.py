@@ -1,0 +1,62 @@
+"""Checkpoint/resume tests (SURVEY.md §5 "Checkpoint / resume"): per-expert
+torch.save maps under checkpoint_dir/<uid>/, checkpoint_last, restore."""
+import torch
+
+from learning_at_home_amd.models import FFNExpert
+from learning_at_home_amd.runtime import ExpertBackend
+from learning_at_home_amd.server import CheckpointSaver, load_experts, store_experts
+from learning_at_home_amd.utils import BatchTensorDescriptor
+
+
+def _make_backend(uid="ffn.0", d=32):
+    expert = FFNExpert(d)
+    opt = torch.optim.SGD(expert.parameters(), lr=0.1)
+    return ExpertBackend(uid, expert, opt, args_schema=(BatchTensorDescriptor.make(d),))
+
+
+class TestCheckpoints:
+    def test_store_load_roundtrip(self, tmp_path):
+        b = _make_backend()
+        # train one async step so optimizer state exists
+        x = torch.randn(4, 32)
+        b.backward(x, torch.randn(4, 32))
+        store_experts({"ffn.0": b}, tmp_path)
+        assert (tmp_path / "ffn.0" / "checkpoint_last.pt").exists()
+        timestamped = [p for p in (tmp_path / "ffn.0").iterdir() if p.name.startswith("checkpoint_2")]
+        assert timestamped, "no timestamped checkpoint written"
+
+        fresh = _make_backend()
+        assert not torch.equal(
+            fresh.expert.fc1.weight.detach(), b.expert.fc1.weight.detach()
+        )
+        restored = load_experts({"ffn.0": fresh}, tmp_path)
+        assert restored == 1
+        assert torch.equal(fresh.expert.fc1.weight.detach(), b.expert.fc1.weight.detach())
+        assert fresh.update_count == b.update_count
+
+    def test_checkpoint_format_is_plain_torch_save(self, tmp_path):
+        """Contract: the file is a torch.save dict with 'model'/'optimizer'
+        keys (byte-level compatibility target per BASELINE.json)."""
+        b = _make_backend()
+        store_experts({"ffn.0": b}, tmp_path)
+        blob = torch.load(tmp_path / "ffn.0" / "checkpoint_last.pt", weights_only=False)
+        assert set(blob.keys()) >= {"model", "optimizer"}
+        assert "fc1.weight" in blob["model"]
+
+    def test_saver_thread(self, tmp_path):
+        b = _make_backend()
+        saver = CheckpointSaver({"ffn.0": b}, tmp_path, update_period=0.2)
+        saver.start()
+        import time
+
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline:
+            if (tmp_path / "ffn.0" / "checkpoint_last.pt").exists():
+                break
+            time.sleep(0.05)
+        saver.stop()
+        assert (tmp_path / "ffn.0" / "checkpoint_last.pt").exists()
+
+    def test_load_missing_is_noop(self, tmp_path):
+        b = _make_backend()
+        assert load_experts({"ffn.0": b}, tmp_path) == 0
