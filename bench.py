@@ -23,6 +23,7 @@ from learning_at_home_amd.models import TransformerDMoE, TransformerDMoEConfig
 from learning_at_home_amd.parallel import ExpertParallelMoE, FusedExpertSGD
 from learning_at_home_amd.averaging import RcclAverager
 from learning_at_home_amd.ops import HAS_HIP
+from learning_at_home_amd.optim import MasterSGD
 
 
 def build_model(cfg: TransformerDMoEConfig, args, device, group=None):
@@ -63,6 +64,9 @@ def main():
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--averaging-period", type=int, default=10)
     p.add_argument("--no-averager", action="store_true")
+    p.add_argument("--fixed-batch", action="store_true",
+                   help="reuse ONE synthetic batch every step (memorization "
+                        "sanity check: loss must fall; NOT a valid bench mode)")
     p.add_argument("--master-weights", action="store_true",
                    help="fp32 master weights for experts (default: bf16 + "
                         "stochastic-rounding SGD, the fast path)")
@@ -109,7 +113,9 @@ def main():
                                               stochastic_rounding=not args.master_weights))
         else:
             expert_opts.append(torch.optim.SGD(block.moe.experts.parameters(), lr=args.lr))
-    trunk_opt = torch.optim.SGD(trunk_params, lr=args.lr)
+    # trunk in bf16 needs fp32 masters: plain SGD on bf16 loses sub-ulp updates
+    trunk_opt = (MasterSGD(trunk_params, lr=args.lr) if use_cuda
+                 else torch.optim.SGD(trunk_params, lr=args.lr))
 
     averager = None
     if world_size > 1 and not args.no_averager:
@@ -117,8 +123,13 @@ def main():
 
     tokens_per_step_per_rank = B * cfg.seq_len
 
+    fixed = None
+    if args.fixed_batch:
+        fixed = torch.randint(0, cfg.vocab_size, (B, cfg.seq_len + 1), device=device)
+
     def make_batch():
-        ids = torch.randint(0, cfg.vocab_size, (B, cfg.seq_len + 1), device=device)
+        ids = fixed if fixed is not None else torch.randint(
+            0, cfg.vocab_size, (B, cfg.seq_len + 1), device=device)
         return ids[:, :-1], ids[:, 1:]
 
     def train_step(step_idx):
@@ -132,9 +143,14 @@ def main():
                 opt.zero_grad(set_to_none=True)
         if averager is not None:
             averager.wait()  # previous round must have landed before we step
+            if getattr(averager, "_round_pending", False):
+                averager._round_pending = False
+                if isinstance(trunk_opt, MasterSGD):
+                    trunk_opt.sync_masters_from_params()  # averaged values win
         trunk_opt.step()
         if averager is not None and (step_idx + 1) % args.averaging_period == 0:
             averager.step(async_op=True)  # overlapped on the side stream
+            averager._round_pending = True
         return loss
 
     def barrier_sync():

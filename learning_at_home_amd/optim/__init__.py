@@ -11,6 +11,7 @@ import torch
 
 from ..averaging import DecentralizedAverager, RcclAverager
 from ..utils import get_logger
+from .master_sgd import MasterSGD
 
 logger = get_logger(__name__)
 
@@ -82,4 +83,4 @@ class DecentralizedSGD:
 
 CollaborativeOptimizer = DecentralizedSGD  # alias (hivemind naming)
 
-__all__ = ["DecentralizedSGD", "CollaborativeOptimizer"]
+__all__ = ["DecentralizedSGD", "CollaborativeOptimizer", "MasterSGD"]

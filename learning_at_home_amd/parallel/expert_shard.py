@@ -175,6 +175,8 @@ class ExpertParallelMoE(nn.Module):
         k: int = 4,
         capacity_factor: float = 1.25,
         aux_loss_coef: float = 0.01,
+        router_noise_std: float = 0.3,
+        z_loss_coef: float = 1e-3,
         group: Optional[object] = None,
         tokens_per_batch_hint: int = 8192,
     ):
@@ -183,6 +185,8 @@ class ExpertParallelMoE(nn.Module):
         self.num_experts, self.k = num_experts, k
         self.capacity_factor = capacity_factor
         self.aux_loss_coef = aux_loss_coef
+        self.router_noise_std = router_noise_std
+        self.z_loss_coef = z_loss_coef
         self.group = group
         self.world_size = dist.get_world_size(group) if dist.is_initialized() else 1
         self.rank = dist.get_rank(group) if dist.is_initialized() else 0
@@ -214,6 +218,11 @@ class ExpertParallelMoE(nn.Module):
         C = self.capacity(T)
 
         logits = self.gate(flat).float()
+        if self.training and self.router_noise_std > 0:
+            # exploration noise (Shazeer-style noisy gating): every expert
+            # keeps receiving tokens early on, preventing rich-get-richer
+            # collapse before the balance loss can act
+            logits = logits + self.router_noise_std * torch.randn_like(logits)
         topk_idx, topk_w, probs = Fops.gating_topk(logits, k)
         with torch.no_grad():
             pos, counts, inv = Fops.assign_slots(topk_idx, E, C)
@@ -244,10 +253,16 @@ class ExpertParallelMoE(nn.Module):
 
         y = Fops.combine(ybuf, topk_idx, pos, topk_w, inv)
 
+        aux = None
         if self.aux_loss_coef > 0:
             f = (counts.float() / max(1, T * k)).detach()
             P = probs.float().mean(dim=0)
-            self.aux_loss = self.aux_loss_coef * E * torch.dot(f.to(P.device), P)
-        else:
-            self.aux_loss = None
+            aux = self.aux_loss_coef * E * torch.dot(f.to(P.device), P)
+        if self.z_loss_coef > 0:
+            # router z-loss (ST-MoE): keeps logit magnitudes bounded so the
+            # gating softmax never saturates into a zero-gradient regime
+            z = torch.logsumexp(logits, dim=-1)
+            z_loss = self.z_loss_coef * (z * z).mean()
+            aux = z_loss if aux is None else aux + z_loss
+        self.aux_loss = aux
         return y.reshape(*lead_shape, D).to(x.dtype)

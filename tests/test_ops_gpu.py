@@ -309,6 +309,7 @@ class TestMoEOnGPU:
         torch.manual_seed(12)
         moe_cpu = ExpertParallelMoE(d_model=256, ffn_dim=512, num_experts=8, k=2,
                                     capacity_factor=8.0, aux_loss_coef=0.0)
+        moe_cpu.eval()
         x = 0.5 * torch.randn(4, 32, 256)
         y_cpu = moe_cpu(x)
         moe_gpu = moe_cpu.to(DEV)

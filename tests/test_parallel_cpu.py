@@ -58,6 +58,7 @@ class TestMoESingleProcess:
         torch.manual_seed(1)
         moe = ExpertParallelMoE(d_model=32, ffn_dim=64, num_experts=4, k=2,
                                 capacity_factor=8.0, aux_loss_coef=0.0)
+        moe.eval()  # exact-match test: no router exploration noise
         x = torch.randn(16, 32)
         y = moe(x)
         y_ref = _manual_moe_reference(x, moe)
@@ -109,6 +110,7 @@ def _ep_worker(rank, world, tmpdir):
     torch.manual_seed(42)
     sharded = ExpertParallelMoE(d_model=32, ffn_dim=64, num_experts=4, k=2,
                                 capacity_factor=16.0, aux_loss_coef=0.0)
+    full.eval(); sharded.eval()  # exact-match: no router noise
     # shard the SAME weights: rank r keeps experts [r*2, r*2+2)
     n_local = 4 // world
     with torch.no_grad():
