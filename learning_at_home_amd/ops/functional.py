@@ -158,7 +158,9 @@ class _Combine(torch.autograd.Function):
 def _survivor_weights(pos, w):
     surv = pos >= 0
     s = (w * surv).sum(-1, keepdim=True)
-    inv_s = torch.where(s > 0, 1.0 / s, torch.zeros_like(s))
+    # clamped renormalization (matches dispatch.hip): bounds the gradient
+    # amplification when capacity drops remove most of the softmax mass
+    inv_s = torch.where(s > 0, 1.0 / s.clamp(min=0.2), torch.zeros_like(s))
     return (w * surv) * inv_s, surv, inv_s
 
 
@@ -184,7 +186,9 @@ def _combine_bwd_torch(dy, ybuf, y, topk_idx, pos, w):
     rows = ybuf.view(-1, D)[(topk_idx.long() * C + pos.clamp(min=0).long()).view(-1)].view(T, k, D)
     dot_yj = torch.einsum("td,tkd->tk", dy.float(), rows.float())
     dot_y = torch.einsum("td,td->t", dy.float(), y.float()).unsqueeze(-1)
-    dw = ((dot_yj - dot_y) * inv_s) * surv
+    s = (w * surv).sum(-1, keepdim=True)
+    clamped = (s > 0) & (s < 0.2)  # in the clamp region d(w/0.2)/dw is diagonal
+    dw = (torch.where(clamped, dot_yj, dot_yj - dot_y) * inv_s) * surv
     return dybuf, dw.float()
 
 

@@ -87,7 +87,10 @@ __global__ void combine_fwd_kernel(const __bf16* __restrict__ ybuf,   // [E, C, 
   float s = 0.f;
   for (int j = 0; j < k; ++j)
     if (pos[(long)t * k + j] >= 0) s += w[(long)t * k + j];
-  float inv_s = (s > 0.f) ? 1.0f / s : 0.f;
+  // clamp the renormalization mass: when capacity drops remove most of the
+  // softmax mass, an unclamped 1/s amplifies combine gradients by up to
+  // 1/w_min and detonates the router (observed at flagship scale)
+  float inv_s = (s > 0.f) ? 1.0f / fmaxf(s, 0.2f) : 0.f;
   for (int c = lane * 4; c < D; c += 64 * 4) {
     float acc[4] = {0.f, 0.f, 0.f, 0.f};
     for (int j = 0; j < k; ++j) {
@@ -124,7 +127,8 @@ __global__ void combine_bwd_kernel(const __bf16* __restrict__ dy,     // [T, D]
   float s = 0.f;
   for (int j = 0; j < k; ++j)
     if (pos[(long)t * k + j] >= 0) s += w[(long)t * k + j];
-  float inv_s = (s > 0.f) ? 1.0f / s : 0.f;
+  float inv_s = (s > 0.f) ? 1.0f / fmaxf(s, 0.2f) : 0.f;  // see combine_fwd
+  bool clamped = (s > 0.f) && (s < 0.2f);  // d(w/0.2)/dw has no quotient term
 
   // <dy, y> once per token
   float dot_dy_y = 0.f;
@@ -160,7 +164,8 @@ __global__ void combine_bwd_kernel(const __bf16* __restrict__ dy,     // [T, D]
     }
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) dot_dy_yj += __shfl_xor(dot_dy_yj, off);
-    if (lane == 0) dw[(long)t * k + j] = (dot_dy_yj - dot_dy_y) * inv_s;
+    if (lane == 0)
+      dw[(long)t * k + j] = (clamped ? dot_dy_yj : (dot_dy_yj - dot_dy_y)) * inv_s;
   }
 }
 
