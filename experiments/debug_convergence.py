@@ -23,18 +23,24 @@ def run(mode: str, steps: int = 120, lr: float = 0.01, scale: str = "tiny"):
         cfg = TransformerDMoEConfig(vocab_size=512, d_model=128, n_heads=4, n_layers=2,
                                     seq_len=64, num_experts=16, k_best=4, ffn_dim=256)
 
+    aux_c = 0.0 if "noaux" in mode else 0.01
+    z_c = 0.0 if "noz" in mode else 1e-3
+    noise = 0.0 if "nonoise" in mode else 0.3
+
     def moe_factory(i, c):
         return ExpertParallelMoE(d_model=c.d_model, ffn_dim=c.ffn_dim,
                                  num_experts=c.num_experts, k=c.k_best,
-                                 capacity_factor=1.5, aux_loss_coef=0.01)
+                                 capacity_factor=1.5, aux_loss_coef=aux_c,
+                                 z_loss_coef=z_c, router_noise_std=noise)
 
     model = TransformerDMoE(cfg, moe_factory).to("cuda:0", torch.bfloat16)
+    sr = "sr" in mode
     expert_opts = []
     for block in model.blocks:
         block.moe.experts.float()
-        block.moe.experts.init_gpu_state(master_weights=(mode != "sr"))
+        block.moe.experts.init_gpu_state(master_weights=not sr)
         expert_opts.append(FusedExpertSGD(block.moe.experts, lr=lr,
-                                          stochastic_rounding=(mode == "sr")))
+                                          stochastic_rounding=sr))
     from learning_at_home_amd.optim import MasterSGD
 
     trunk = [p for n, p in model.named_parameters() if ".moe.experts." not in n]
@@ -55,8 +61,13 @@ def run(mode: str, steps: int = 120, lr: float = 0.01, scale: str = "tiny"):
         topt.step()
         if step % 10 == 0 or step == steps - 1:
             counts = model.blocks[0].moe.last_counts
+            gate = model.blocks[0].moe.gate.weight
+            gnorm = float(gate.grad.float().norm()) if gate.grad is not None else -1
+            with torch.no_grad():
+                lmax = float(gate.float().abs().max())
             print(f"step {step:3d} ce {float(ce):.4f} aux {float(aux):.4f} "
-                  f"cmax {int(counts.max())} cmin {int(counts.min())}", flush=True)
+                  f"cmax {int(counts.max())} cmin {int(counts.min())} "
+                  f"|gate.grad| {gnorm:.3e} |gate|max {lmax:.3e}", flush=True)
 
 
 if __name__ == "__main__":

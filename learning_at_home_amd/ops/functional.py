@@ -45,7 +45,11 @@ class _GatingTopK(torch.autograd.Function):
         probs, topk_idx, topk_w = ctx.saved_tensors
         aux_g = None
         if d_probs is not None:
-            aux_g = d_probs.sum(dim=0).float().contiguous()  # token-uniform reduction
+            # the kernel applies the per-token softmax jacobian with ONE row
+            # vector g: for token-uniform d_probs the correct row is the MEAN
+            # over tokens (sum would overcount by T — a 2000x gate-gradient
+            # blowup observed at flagship scale)
+            aux_g = d_probs.float().mean(dim=0).contiguous()
         if d_topk_w is None:
             d_topk_w = torch.zeros_like(topk_w)
         d_logits = _C.gating_topk_bwd(probs, topk_idx, topk_w,

@@ -13,7 +13,7 @@ void lah_batched_gemm_tn(const void*, const void*, void*, int, int, int, int, in
 void lah_gating_topk_fwd(const void*, void*, void*, void*, int, int, int, hipStream_t);
 void lah_gating_topk_bwd(const void*, const void*, const void*, const void*, const void*,
                          void*, int, int, int, hipStream_t);
-void lah_assign_slots(const void*, void*, void*, void*, int, int, int, hipStream_t);
+void lah_assign_slots(const void*, void*, void*, void*, int, int, int, int, hipStream_t);
 void lah_dispatch_gather(const void*, const void*, void*, int, int, int, int, hipStream_t);
 void lah_dispatch_bwd(const void*, const void*, const void*, void*, int, int, int, int,
                       hipStream_t);
@@ -112,7 +112,7 @@ std::vector<torch::Tensor> assign_slots(torch::Tensor topk_idx, int64_t E, int64
   auto counts = torch::zeros({E}, topk_idx.options());
   auto inv = torch::full({E, C}, -1, topk_idx.options());
   lah_assign_slots(topk_idx.data_ptr(), pos.data_ptr(), counts.data_ptr(), inv.data_ptr(),
-                   T, k, (int)C, cur_stream());
+                   T, k, (int)C, (int)E, cur_stream());
   return {pos, counts, inv};
 }
 

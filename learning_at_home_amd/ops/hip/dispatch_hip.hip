@@ -16,10 +16,14 @@ __global__ void assign_slots_kernel(const int* __restrict__ topk_idx,  // [T, k]
                                     int* __restrict__ pos,             // [T, k]
                                     int* __restrict__ counts,          // [E] zeroed
                                     int* __restrict__ inv,             // [E, C] (-1 fill)
-                                    int total, int C) {
+                                    int total, int C, int E) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= total) return;
   int e = topk_idx[i];
+  if (e < 0 || e >= E) {  // defensive: corrupt index must not become OOB
+    pos[i] = -1;
+    return;
+  }
   int p = atomicAdd(&counts[e], 1);
   if (p < C) {
     pos[i] = p;
@@ -182,10 +186,11 @@ __global__ void zero_padding_kernel(const int* __restrict__ inv, __bf16* __restr
 
 // ---- launchers -------------------------------------------------------------
 extern "C" void lah_assign_slots(const void* topk_idx, void* pos, void* counts,
-                                 void* inv, int T, int k, int C, hipStream_t stream) {
+                                 void* inv, int T, int k, int C, int E,
+                                 hipStream_t stream) {
   int total = T * k;
   hipLaunchKernelGGL(assign_slots_kernel, dim3(cdiv(total, 256)), dim3(256), 0, stream,
-                     (const int*)topk_idx, (int*)pos, (int*)counts, (int*)inv, total, C);
+                     (const int*)topk_idx, (int*)pos, (int*)counts, (int*)inv, total, C, E);
 }
 
 extern "C" void lah_dispatch_gather(const void* x, const void* inv, void* xbuf, int E,

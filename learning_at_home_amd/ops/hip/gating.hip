@@ -74,8 +74,12 @@ __global__ void gating_topk_fwd_kernel(const float* __restrict__ logits,  // [T,
 #pragma unroll
     for (int i = 0; i < EPL; ++i) {
       bool taken = (sel_mask >> i) & 1u;
-      if (!taken && lv[i] > best) { best = lv[i]; bi = i; }
+      // NaN-safe: a NaN logit never wins, and bi falls back to the first
+      // untaken entry so sel_idx is ALWAYS a valid expert (a negative index
+      // here would turn into an out-of-bounds atomic in assign_slots)
+      if (!taken && (bi < 0 || lv[i] > best)) { best = lv[i]; bi = i; }
     }
+    if (bi < 0) bi = 0;
     // argmax across lanes: pack (value, lane) and reduce
     float bv = best;
     int bl = lane;

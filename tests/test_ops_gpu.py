@@ -127,6 +127,33 @@ class TestGating:
         assert_close_bf16(d_logits, logits.grad, rtol=2e-2, what="gating bwd")
 
 
+class TestGatingFunctionIntegration:
+    def test_full_function_grads_match_cpu_autograd(self):
+        """gating_topk (Function wiring incl. the aux d_probs reduction) vs
+        the pure-torch CPU path on the same logits — catches wiring-level
+        scale bugs that per-kernel tests cannot see."""
+        torch.manual_seed(30)
+        T, E, k = 512, 256, 4
+        logits_gpu = torch.randn(T, E, device=DEV, requires_grad=True)
+        logits_cpu = logits_gpu.detach().cpu().requires_grad_(True)
+        d_w = torch.randn(T, k)
+        aux_vec = torch.rand(E) * 0.1
+
+        idx_g, w_g, p_g = Fops.gating_topk(logits_gpu, k)
+        # aux-style loss: coef-weighted mean prob per expert (token-uniform d_probs)
+        loss_g = (w_g * d_w.to(DEV)).sum() + (p_g.float().mean(0) * aux_vec.to(DEV)).sum()
+        loss_g.backward()
+
+        idx_c, w_c, p_c = Fops.gating_topk(logits_cpu, k)
+        # align slot order with the GPU kernel's selection order
+        order = idx_g.long().cpu()
+        w_c_ordered = torch.softmax(logits_cpu.gather(1, order), dim=-1)
+        loss_c = (w_c_ordered * d_w).sum() + (p_c.mean(0) * aux_vec).sum()
+        loss_c.backward()
+        assert_close_bf16(logits_gpu.grad, logits_cpu.grad, rtol=2e-2,
+                          what="gating Function grads (aux scale)")
+
+
 class TestDispatchCombine:
     def _setup(self, T=300, E=16, k=4, C=128, D=256):
         torch.manual_seed(6)
