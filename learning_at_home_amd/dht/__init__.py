@@ -62,9 +62,20 @@ class DHT(mp.Process):
         self._parent_conn, self._child_conn = mp.Pipe()
         self._port = mp.Value("i", 0)
         self.ready = mp.Event()
-        self._lock = threading.Lock()
+        self._lock_obj: Optional[threading.Lock] = None  # lazy: keeps DHT spawn-picklable
         if start:
             self.run_in_background()
+
+    @property
+    def _lock(self) -> threading.Lock:
+        if self._lock_obj is None:
+            self._lock_obj = threading.Lock()
+        return self._lock_obj
+
+    def __getstate__(self):
+        state = self.__dict__.copy()
+        state["_lock_obj"] = None
+        return state
 
     # ---- lifecycle ----------------------------------------------------
     def run_in_background(self, await_ready: bool = True, timeout: float = 15.0) -> None:

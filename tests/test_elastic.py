@@ -1,0 +1,130 @@
+"""Elastic EP membership tests (BASELINE.json config #5: drop/rejoin an
+expert rank mid-epoch with DHT re-discovery; SURVEY.md §5 generation-fenced
+communicator rebuild). Runs over gloo on localhost — the same code path
+drives RCCL on GPUs."""
+import multiprocessing as mp
+import os
+import time
+
+import pytest
+import torch
+import torch.distributed as dist
+
+from learning_at_home_amd import DHT
+
+LEASE = 2.0
+
+
+def _worker(slot, seed_endpoint, results, rejoiner=False):
+    from learning_at_home_amd.parallel.elastic import ElasticGroup
+
+    dht = DHT(initial_peers=[tuple(seed_endpoint)], start=True)
+    group = ElasticGroup(dht, slot, max_slots=4, lease=LEASE, backend="gloo", timeout=30.0)
+    group.start_announcer()
+    try:
+        if not rejoiner:
+            # phase 1: all three form a group
+            group.form_group(expected=3, deadline=60.0)
+            t = torch.tensor([float(slot)])
+            dist.all_reduce(t)
+            assert t.item() == 3.0, f"phase1 allreduce got {t.item()}"
+            results.put(("phase1", slot, group.gen, group.world_size))
+
+            if slot == 2:
+                group.leave()  # die mid-epoch
+                dht.shutdown()
+                return
+
+            # phase 2: survivors wait out the lease, re-form as a pair
+            deadline = time.monotonic() + 60
+            while time.monotonic() < deadline:
+                if 2 not in group.alive_slots():
+                    break
+                time.sleep(0.3)
+            group.form_group(expected=2, deadline=60.0)
+            t = torch.tensor([float(slot)])
+            dist.all_reduce(t)
+            assert t.item() == 1.0, f"phase2 allreduce got {t.item()}"
+            results.put(("phase2", slot, group.gen, group.world_size))
+
+            # phase 3: slot 2 rejoins (new process, same slot)
+            deadline = time.monotonic() + 90
+            while time.monotonic() < deadline:
+                if 2 in group.alive_slots():
+                    break
+                time.sleep(0.3)
+            group.form_group(expected=3, deadline=90.0)
+            t = torch.tensor([float(slot)])
+            dist.all_reduce(t)
+            assert t.item() == 3.0, f"phase3 allreduce got {t.item()}"
+            results.put(("phase3", slot, group.gen, group.world_size))
+        else:
+            group.form_group(expected=3, deadline=90.0)
+            t = torch.tensor([float(slot)])
+            dist.all_reduce(t)
+            assert t.item() == 3.0
+            results.put(("phase3", slot, group.gen, group.world_size))
+    finally:
+        try:
+            group.leave()
+        except Exception:  # noqa: BLE001
+            pass
+        dht.shutdown()
+
+
+@pytest.mark.timeout(300)
+def test_drop_and_rejoin_rank():
+    ctx = mp.get_context("spawn")
+    seed = DHT(start=True)
+    results = ctx.Queue()
+    procs = [
+        ctx.Process(target=_worker, args=(slot, list(seed.endpoint), results))
+        for slot in range(3)
+    ]
+    for p in procs:
+        p.start()
+    try:
+        # collect phase1 from all three
+        seen = {}
+        deadline = time.monotonic() + 120
+        while len([k for k in seen if k[0] == "phase1"]) < 3 and time.monotonic() < deadline:
+            try:
+                tag, slot, gen, world = results.get(timeout=5)
+                seen[(tag, slot)] = (gen, world)
+            except Exception:  # noqa: BLE001
+                pass
+        assert len([k for k in seen if k[0] == "phase1"]) == 3, f"phase1 incomplete: {seen}"
+        assert all(w == 3 for k, (g, w) in seen.items() if k[0] == "phase1")
+
+        # collect phase2 from survivors (0 and 1)
+        while len([k for k in seen if k[0] == "phase2"]) < 2 and time.monotonic() < deadline:
+            try:
+                tag, slot, gen, world = results.get(timeout=5)
+                seen[(tag, slot)] = (gen, world)
+            except Exception:  # noqa: BLE001
+                pass
+        phase2 = {k: v for k, v in seen.items() if k[0] == "phase2"}
+        assert len(phase2) == 2, f"phase2 incomplete: {seen}"
+        assert all(w == 2 for g, w in phase2.values()), "survivors did not re-form as a pair"
+
+        # now start the rejoiner for slot 2
+        rejoin = ctx.Process(target=_worker, args=(2, list(seed.endpoint), results, True))
+        rejoin.start()
+        procs.append(rejoin)
+
+        deadline = time.monotonic() + 150
+        while len([k for k in seen if k[0] == "phase3"]) < 3 and time.monotonic() < deadline:
+            try:
+                tag, slot, gen, world = results.get(timeout=5)
+                seen[(tag, slot)] = (gen, world)
+            except Exception:  # noqa: BLE001
+                pass
+        phase3 = {k: v for k, v in seen.items() if k[0] == "phase3"}
+        assert len(phase3) == 3, f"phase3 incomplete: {seen}"
+        assert all(w == 3 for g, w in phase3.values()), "rejoin did not restore world=3"
+    finally:
+        for p in procs:
+            p.join(timeout=30)
+            if p.is_alive():
+                p.terminate()
+        seed.shutdown()
