@@ -67,9 +67,11 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
                             int M, int N, int K) {
   constexpr int BM = WM * FM * 16;
   constexpr int BN = WN * FN * 16;
-  __shared__ __bf16 smem[2 * (BM + BN) * BK];
+  __shared__ __attribute__((aligned(16))) __bf16 smem[2 * (BM + BN) * BK];
   const int HALF = (BM + BN) * BK;
 
+  // (XCD-aware remap was A/B-tested here and REGRESSED: the multi-dim grid
+  // dispatch order does not match the 1-D b%8 placement the remap assumes)
   const int e = blockIdx.z;
   const int n0 = blockIdx.x * BN;
   const int m0 = blockIdx.y * BM;
@@ -131,36 +133,62 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
     cur ^= 1;
   }
 
-  // epilogue: C/D fragment mapping for 16x16x32: col = lane&15,
-  // row = (lane>>4)*4 + reg  (cdna_hip_programming.md §3)
-  __bf16* Ce = C + (long)e * M * N;
-  __bf16* De = (EPI == EPI_BIAS_GELU || EPI == EPI_GELU_GRAD)
-                   ? DER + (long)e * M * N : nullptr;
-  const float* be = (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) ? bias + (long)e * N : nullptr;
+  // epilogue through LDS: scatter the C/D fragments (col = lane&15,
+  // row = (lane>>4)*4 + reg — guide §3) into an fp32 LDS image, then emit
+  // COALESCED bf16x8 stores (the direct scalar-store epilogue costs ~2 µs
+  // per workgroup, ~40% of a K=1024 tile's time). The fp32 tile reuses the
+  // staging LDS (BM*BN*4 <= 2*(BM+BN)*BK*2 for both instantiated shapes).
+  static_assert(BM * BN * 4 <= 2 * (BM + BN) * BK * 2, "epilogue tile must fit LDS");
+  float* ctile = (float*)smem;
+  __syncthreads();  // staging reads done; smem is free
 #pragma unroll
   for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
     for (int ni = 0; ni < FN; ++ni) {
-      int col = n0 + wc * FN * 16 + ni * 16 + l15;
+      int col = wc * FN * 16 + ni * 16 + l15;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int row = m0 + wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
-        if (row >= M) continue;
-        long off = (long)row * N + col;
-        float v = acc[mi][ni][r];
+        int row = wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
+        ctile[row * BN + col] = acc[mi][ni][r];
+      }
+    }
+  }
+  __syncthreads();
+
+  __bf16* Ce = C + (long)e * M * N;
+  __bf16* De = (EPI == EPI_BIAS_GELU || EPI == EPI_GELU_GRAD)
+                   ? DER + (long)e * M * N : nullptr;
+  const float* be = (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) ? bias + (long)e * N : nullptr;
+  // each thread: 16 consecutive cols of a row; 8 threads per row;
+  // NTHREADS/8 = 32 rows per pass
+  const int ecol = (tid & 7) * 16;
+  for (int rr = tid >> 3; rr < BM; rr += NTHREADS >> 3) {
+    int row = m0 + rr;
+    if (row >= M) continue;
+    long goff = (long)row * N + n0 + ecol;
+    bf16x8 outv[2], derv[2];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 dv{};
+      if (EPI == EPI_GELU_GRAD) dv = *(const bf16x8*)(De + goff + h * 8);
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        float v = ctile[rr * BN + ecol + h * 8 + q];
         if (EPI == EPI_BIAS) {
-          Ce[off] = f2bf(v + be[col]);
+          outv[h][q] = f2bf(v + be[n0 + ecol + h * 8 + q]);
         } else if (EPI == EPI_BIAS_GELU) {
           float g, dg;
-          gelu_both(v + be[col], &g, &dg);
-          Ce[off] = f2bf(g);
-          De[off] = f2bf(dg);
+          gelu_both(v + be[n0 + ecol + h * 8 + q], &g, &dg);
+          outv[h][q] = f2bf(g);
+          derv[h][q] = f2bf(dg);
         } else if (EPI == EPI_GELU_GRAD) {
-          Ce[off] = f2bf(v * bf2f(De[off]));
+          outv[h][q] = f2bf(v * bf2f(dv[q]));
         } else {
-          Ce[off] = f2bf(v);
+          outv[h][q] = f2bf(v);
         }
       }
+      *(bf16x8*)(Ce + goff + h * 8) = outv[h];
+      if (EPI == EPI_BIAS_GELU) *(bf16x8*)(De + goff + h * 8) = derv[h];
     }
   }
 }

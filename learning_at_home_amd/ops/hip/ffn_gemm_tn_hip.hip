@@ -43,7 +43,7 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
                             const __bf16* __restrict__ B,  // [E, M, J]
                             float* __restrict__ C,         // [E, I, J]
                             int M, int I, int J) {
-  __shared__ __bf16 smem[2 * (TBI + TBJ) * TBK];
+  __shared__ __attribute__((aligned(16))) __bf16 smem[2 * (TBI + TBJ) * TBK];
   const int HALF = (TBI + TBJ) * TBK;
 
   const int e = blockIdx.z;
@@ -104,6 +104,9 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
     cur ^= 1;
   }
 
+  // direct fragment-mapped epilogue (an LDS-staged variant was A/B-tested
+  // and LOST here: the TN kernel is HBM-bound, and the fp32 LDS roundtrip +
+  // 2 barriers cost more than the store coalescing gains)
   float* Ce = C + (long)e * I * J;
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi) {
