@@ -94,18 +94,29 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
   int cur = 0;
   stage_tile<BM, true>(Ae, smem, m0, M, K, 0, tid);
   stage_tile<BN, false>(Be, smem + BM * BK, n0, N, K, 0, tid);
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-  __syncthreads();
 
   const int l15 = lane & 15;
   const int lk8 = (lane >> 4) * 8;
+  // glds instructions per thread per stage — the counted-vmcnt quantum
+  constexpr int GLDS_PER_STAGE = BM / 32 + BN / 32;
 
   for (int t = 0; t < nk; ++t) {
     if (t + 1 < nk) {
       stage_tile<BM, true>(Ae, smem + (cur ^ 1) * HALF, m0, M, K, (t + 1) * BK, tid);
       stage_tile<BN, false>(Be, smem + (cur ^ 1) * HALF + BM * BK, n0, N, K,
                             (t + 1) * BK, tid);
+      // counted wait: stage t landed, stage t+1 stays IN FLIGHT across the
+      // barrier (raw s_barrier — __syncthreads would emit vmcnt(0) and drain
+      // the prefetch; guide §5 "pipelining across barriers" / T3+T4)
+      if constexpr (GLDS_PER_STAGE == 8)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     }
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
     const __bf16* la = smem + cur * HALF;
     const __bf16* lb = smem + cur * HALF + BM * BK;
 #pragma unroll
@@ -129,9 +140,9 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[mi], bfrag[ni],
                                                                 acc[mi][ni], 0, 0, 0);
     }
-    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    __syncthreads();
-    cur ^= 1;
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_barrier();  // all waves done READING buf[cur] before
+    cur ^= 1;                      // the next iteration's glds overwrite it
   }
 
   // epilogue through LDS: scatter the C/D fragments (col = lane&15,
