@@ -70,6 +70,8 @@ def main():
     p.add_argument("--master-weights", action="store_true",
                    help="fp32 master weights for experts (default: bf16 + "
                         "stochastic-rounding SGD, the fast path)")
+    p.add_argument("--metrics", action="store_true",
+                   help="print per-stage timers + expert-load stats to stderr")
     args = p.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -132,22 +134,33 @@ def main():
             0, cfg.vocab_size, (B, cfg.seq_len + 1), device=device)
         return ids[:, :-1], ids[:, 1:]
 
+    from learning_at_home_amd.utils.monitor import StageTimers, expert_load_stats
+
+    timers = StageTimers() if args.metrics else None
+
     def train_step(step_idx):
+        from contextlib import nullcontext
+
+        tm = timers if timers is not None else (lambda s: nullcontext())
         inputs, labels = make_batch()
-        loss = model.loss(inputs, labels)
+        with tm("forward"):
+            loss = model.loss(inputs, labels)
         trunk_opt.zero_grad(set_to_none=True)
-        loss.backward()
-        for opt in expert_opts:
-            opt.step()  # async per-shard SGD (fused on GPU)
-            if isinstance(opt, torch.optim.SGD):
-                opt.zero_grad(set_to_none=True)
+        with tm("backward"):
+            loss.backward()
+        with tm("expert_opt"):
+            for opt in expert_opts:
+                opt.step()  # async per-shard SGD (fused on GPU)
+                if isinstance(opt, torch.optim.SGD):
+                    opt.zero_grad(set_to_none=True)
         if averager is not None:
             averager.wait()  # previous round must have landed before we step
             if getattr(averager, "_round_pending", False):
                 averager._round_pending = False
                 if isinstance(trunk_opt, MasterSGD):
                     trunk_opt.sync_masters_from_params()  # averaged values win
-        trunk_opt.step()
+        with tm("trunk_opt"):
+            trunk_opt.step()
         if averager is not None and (step_idx + 1) % args.averaging_period == 0:
             averager.step(async_op=True)  # overlapped on the side stream
             averager._round_pending = True
@@ -202,6 +215,16 @@ def main():
                 "last_loss": round(float(last_loss), 4) if last_loss is not None else None,
             },
         }))
+
+    if args.metrics and rank == 0:
+        import sys
+
+        print("[metrics] stages:", timers.report(), file=sys.stderr)
+        counts = model.blocks[0].moe.last_counts
+        if counts is not None:
+            cap = model.blocks[0].moe.capacity(tokens_per_step_per_rank)
+            print("[metrics] expert load (layer 0):",
+                  json.dumps(expert_load_stats(counts.cpu(), cap)), file=sys.stderr)
 
     if world_size > 1:
         dist.destroy_process_group()

@@ -12,19 +12,19 @@ import torch.distributed as dist
 
 from learning_at_home_amd import DHT
 
-LEASE = 2.0
+LEASE = 3.0
 
 
 def _worker(slot, seed_endpoint, results, rejoiner=False):
     from learning_at_home_amd.parallel.elastic import ElasticGroup
 
     dht = DHT(initial_peers=[tuple(seed_endpoint)], start=True)
-    group = ElasticGroup(dht, slot, max_slots=4, lease=LEASE, backend="gloo", timeout=30.0)
+    group = ElasticGroup(dht, slot, max_slots=4, lease=LEASE, backend="gloo", timeout=60.0)
     group.start_announcer()
     try:
         if not rejoiner:
             # phase 1: all three form a group
-            group.form_group(expected=3, deadline=60.0)
+            group.form_group(expected=3, deadline=120.0)
             t = torch.tensor([float(slot)])
             dist.all_reduce(t)
             assert t.item() == 3.0, f"phase1 allreduce got {t.item()}"
@@ -41,7 +41,7 @@ def _worker(slot, seed_endpoint, results, rejoiner=False):
                 if 2 not in group.alive_slots():
                     break
                 time.sleep(0.3)
-            group.form_group(expected=2, deadline=60.0)
+            group.form_group(expected=2, deadline=120.0)
             t = torch.tensor([float(slot)])
             dist.all_reduce(t)
             assert t.item() == 1.0, f"phase2 allreduce got {t.item()}"
@@ -53,13 +53,13 @@ def _worker(slot, seed_endpoint, results, rejoiner=False):
                 if 2 in group.alive_slots():
                     break
                 time.sleep(0.3)
-            group.form_group(expected=3, deadline=90.0)
+            group.form_group(expected=3, deadline=150.0)
             t = torch.tensor([float(slot)])
             dist.all_reduce(t)
             assert t.item() == 3.0, f"phase3 allreduce got {t.item()}"
             results.put(("phase3", slot, group.gen, group.world_size))
         else:
-            group.form_group(expected=3, deadline=90.0)
+            group.form_group(expected=3, deadline=150.0)
             t = torch.tensor([float(slot)])
             dist.all_reduce(t)
             assert t.item() == 3.0
