@@ -18,6 +18,7 @@
 // 64x128 (4 waves of 64x32) otherwise — capacity buffers are padded to 64
 // rows, so a 192-row expert batch runs as 3 full 64-tiles with zero waste.
 #include "common.h"
+#include <cstdlib>
 
 #define BK 64
 #define NTHREADS 256
@@ -30,12 +31,12 @@
 // swizzled LDS element offset of 16-byte chunk c in row `row` (guide T2)
 #define LDS_CHUNK(row, c) (((row) * 8 + ((c) ^ ((row) & 7))) * 8)
 
-template <int ROWS, bool CLAMP>
+template <int ROWS, bool CLAMP, int THREADS = NTHREADS>
 LAH_DEVINL void stage_tile(const __bf16* __restrict__ src, __bf16* lds, int row0,
                            int max_row, int ld, int k0, int tid) {
 #pragma unroll
-  for (int i = 0; i < (ROWS * BK) / (NTHREADS * 8); ++i) {
-    int idx = i * NTHREADS + tid;       // 16-byte chunk index
+  for (int i = 0; i < (ROWS * BK) / (THREADS * 8); ++i) {
+    int idx = i * THREADS + tid;        // 16-byte chunk index
     int row = idx >> 3;                 // BK=64 -> 8 chunks per row
     int c8 = (idx & 7) ^ (row & 7);     // inverse swizzle on the source side
     int grow = row0 + row;
@@ -54,11 +55,55 @@ LAH_DEVINL void gelu_both(float x, float* g, float* dg) {
   *dg = 0.5f * (1.0f + t) + 0.5f * x * (1.0f - t * t) * du;
 }
 
+// coalesced bf16x8 store pass: ERB rows of the fp32 LDS tile -> global
+template <int EPI, int BN, int THREADS>
+LAH_DEVINL void emit_rows(const float* ctile, __bf16* Ce, __bf16* De, const float* be,
+                          int row0, int n0, int M, int N, int nrows, int tid, int /*unused*/) {
+  constexpr int TPR = BN / 16;  // threads per row (16 cols x bf16x8 pair each)
+  const int ecol = (tid % TPR) * 16;
+  // bias is row-invariant: load this thread's 16-wide chunk ONCE
+  float bv[16];
+  if (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) {
+#pragma unroll
+    for (int q = 0; q < 16; q += 4)
+      *(f32x4*)(bv + q) = *(const f32x4*)(be + n0 + ecol + q);
+  }
+  for (int rr = tid / TPR; rr < nrows; rr += THREADS / TPR) {
+    int row = row0 + rr;
+    if (row >= M) continue;
+    long goff = (long)row * N + n0 + ecol;
+    bf16x8 outv[2], derv[2];
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      bf16x8 dv{};
+      if (EPI == EPI_GELU_GRAD) dv = *(const bf16x8*)(De + goff + h * 8);
+#pragma unroll
+      for (int q = 0; q < 8; ++q) {
+        float v = ctile[rr * BN + ecol + h * 8 + q];
+        if (EPI == EPI_BIAS) {
+          outv[h][q] = f2bf(v + bv[h * 8 + q]);
+        } else if (EPI == EPI_BIAS_GELU) {
+          float g, dg;
+          gelu_both(v + bv[h * 8 + q], &g, &dg);
+          outv[h][q] = f2bf(g);
+          derv[h][q] = f2bf(dg);
+        } else if (EPI == EPI_GELU_GRAD) {
+          outv[h][q] = f2bf(v * bf2f(dv[q]));
+        } else {
+          outv[h][q] = f2bf(v);
+        }
+      }
+      *(bf16x8*)(Ce + goff + h * 8) = outv[h];
+      if (EPI == EPI_BIAS_GELU) *(bf16x8*)(De + goff + h * 8) = derv[h];
+    }
+  }
+}
+
 // WM x WN waves, each computing FM x FN fragments of 16x16:
 // BM = WM*FM*16, BN = WN*FN*16. Instantiated as <EPI, 2,2,4,4> (128x128)
 // and <EPI, 1,4,4,2> (64x128).
 template <int EPI, int WM, int WN, int FM, int FN>
-__launch_bounds__(NTHREADS, 2) __global__
+__launch_bounds__(WM * WN * 64, (WM * WN > 4) ? 1 : 2) __global__
 void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
                             const __bf16* __restrict__ B,  // [E, N, K]
                             __bf16* __restrict__ C,        // [E, M, N]
@@ -67,6 +112,7 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
                             int M, int N, int K) {
   constexpr int BM = WM * FM * 16;
   constexpr int BN = WN * FN * 16;
+  constexpr int THREADS = WM * WN * 64;
   __shared__ __attribute__((aligned(16))) __bf16 smem[2 * (BM + BN) * BK];
   const int HALF = (BM + BN) * BK;
 
@@ -91,18 +137,18 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
 
   const int nk = K / BK;
   int cur = 0;
-  stage_tile<BM, true>(Ae, smem, m0, M, K, 0, tid);
-  stage_tile<BN, false>(Be, smem + BM * BK, n0, N, K, 0, tid);
+  stage_tile<BM, true, THREADS>(Ae, smem, m0, M, K, 0, tid);
+  stage_tile<BN, false, THREADS>(Be, smem + BM * BK, n0, N, K, 0, tid);
 
   const int l15 = lane & 15;
   const int lk8 = (lane >> 4) * 8;
   // glds instructions per thread per stage — the counted-vmcnt quantum
-  constexpr int GLDS_PER_STAGE = BM / 32 + BN / 32;
+  constexpr int GLDS_PER_STAGE = (BM * BK) / (THREADS * 8) + (BN * BK) / (THREADS * 8);
 
   for (int t = 0; t < nk; ++t) {
     if (t + 1 < nk) {
-      stage_tile<BM, true>(Ae, smem + (cur ^ 1) * HALF, m0, M, K, (t + 1) * BK, tid);
-      stage_tile<BN, false>(Be, smem + (cur ^ 1) * HALF + BM * BK, n0, N, K,
+      stage_tile<BM, true, THREADS>(Ae, smem + (cur ^ 1) * HALF, m0, M, K, (t + 1) * BK, tid);
+      stage_tile<BN, false, THREADS>(Be, smem + (cur ^ 1) * HALF + BM * BK, n0, N, K,
                             (t + 1) * BK, tid);
       // counted wait: stage t landed, stage t+1 stays IN FLIGHT across the
       // barrier (raw s_barrier — __syncthreads would emit vmcnt(0) and drain
@@ -149,72 +195,65 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
   // COALESCED bf16x8 stores (the direct scalar-store epilogue costs ~2 µs
   // per workgroup, ~40% of a K=1024 tile's time). The fp32 tile reuses the
   // staging LDS (BM*BN*4 <= 2*(BM+BN)*BK*2 for both instantiated shapes).
-  static_assert(BM * BN * 4 <= 2 * (BM + BN) * BK * 2, "epilogue tile must fit LDS");
+  // rows per epilogue pass: limited by the fp32 LDS image (multi-pass for
+  // the 256x256 tile whose full image would need 256 KiB)
+  constexpr int LDS_BYTES = 2 * (BM + BN) * BK * 2;
+  constexpr int ERB = (LDS_BYTES / (BN * 4)) < BM ? (LDS_BYTES / (BN * 4)) : BM;
+  static_assert(ERB >= 16 && BM % ERB == 0, "bad epilogue row split");
   float* ctile = (float*)smem;
-  __syncthreads();  // staging reads done; smem is free
+  const int ncol = (tid & 7) * 16;
 #pragma unroll
-  for (int mi = 0; mi < FM; ++mi) {
+  for (int ep = 0; ep < BM / ERB; ++ep) {
+    __syncthreads();  // staging/previous-pass reads done; smem is free
 #pragma unroll
-    for (int ni = 0; ni < FN; ++ni) {
-      int col = wc * FN * 16 + ni * 16 + l15;
+    for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int row = wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
-        ctile[row * BN + col] = acc[mi][ni][r];
-      }
-    }
-  }
-  __syncthreads();
-
-  __bf16* Ce = C + (long)e * M * N;
-  __bf16* De = (EPI == EPI_BIAS_GELU || EPI == EPI_GELU_GRAD)
-                   ? DER + (long)e * M * N : nullptr;
-  const float* be = (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) ? bias + (long)e * N : nullptr;
-  // each thread: 16 consecutive cols of a row; 8 threads per row;
-  // NTHREADS/8 = 32 rows per pass
-  const int ecol = (tid & 7) * 16;
-  for (int rr = tid >> 3; rr < BM; rr += NTHREADS >> 3) {
-    int row = m0 + rr;
-    if (row >= M) continue;
-    long goff = (long)row * N + n0 + ecol;
-    bf16x8 outv[2], derv[2];
+      for (int ni = 0; ni < FN; ++ni) {
+        int col = wc * FN * 16 + ni * 16 + l15;
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      bf16x8 dv{};
-      if (EPI == EPI_GELU_GRAD) dv = *(const bf16x8*)(De + goff + h * 8);
-#pragma unroll
-      for (int q = 0; q < 8; ++q) {
-        float v = ctile[rr * BN + ecol + h * 8 + q];
-        if (EPI == EPI_BIAS) {
-          outv[h][q] = f2bf(v + be[n0 + ecol + h * 8 + q]);
-        } else if (EPI == EPI_BIAS_GELU) {
-          float g, dg;
-          gelu_both(v + be[n0 + ecol + h * 8 + q], &g, &dg);
-          outv[h][q] = f2bf(g);
-          derv[h][q] = f2bf(dg);
-        } else if (EPI == EPI_GELU_GRAD) {
-          outv[h][q] = f2bf(v * bf2f(dv[q]));
-        } else {
-          outv[h][q] = f2bf(v);
+        for (int r = 0; r < 4; ++r) {
+          int row = wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
+          if (row / ERB == ep) ctile[(row % ERB) * BN + col] = acc[mi][ni][r];
         }
       }
-      *(bf16x8*)(Ce + goff + h * 8) = outv[h];
-      if (EPI == EPI_BIAS_GELU) *(bf16x8*)(De + goff + h * 8) = derv[h];
     }
+    __syncthreads();
+
+    __bf16* Ce = C + (long)e * M * N;
+    __bf16* De = (EPI == EPI_BIAS_GELU || EPI == EPI_GELU_GRAD)
+                     ? DER + (long)e * M * N : nullptr;
+    const float* be = (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) ? bias + (long)e * N : nullptr;
+    emit_rows<EPI, BN, THREADS>(ctile, Ce, De, be, m0 + ep * ERB, n0, M, N, ERB, tid, ncol);
   }
+}
+
+
+// runtime switch for the big-tile variant (A/B tuning; see bench_kernels.py)
+static int lah_use_256_tile() {
+  static int v = -1;
+  if (v < 0) {
+    const char* env = getenv("LAH_GEMM_256");
+    v = (env == nullptr || env[0] != '0') ? 1 : 0;
+  }
+  return v;
 }
 
 template <int EPI>
 static void launch_nt(const void* A, const void* B, void* C, const void* bias,
                       void* DER, int E, int M, int N, int K, hipStream_t stream) {
-  dim3 block(NTHREADS);
-  if (M % 128 == 0) {
-    dim3 grid(N / 128, M / 128, E);
+  if (M % 256 == 0 && N % 256 == 0 && lah_use_256_tile()) {
+    // 256x256 tile, 8 waves, 128 KiB LDS, 1 block/CU (guide: high-tile regime)
+    dim3 grid(N / 256, M / 256, E), block(512);
+    hipLaunchKernelGGL((batched_gemm_nt_kernel<EPI, 2, 4, 8, 4>), grid, block, 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
+                       (const float*)bias, (__bf16*)DER, M, N, K);
+  } else if (M % 128 == 0) {
+    dim3 grid(N / 128, M / 128, E), block(256);
     hipLaunchKernelGGL((batched_gemm_nt_kernel<EPI, 2, 2, 4, 4>), grid, block, 0, stream,
                        (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
                        (const float*)bias, (__bf16*)DER, M, N, K);
   } else {  // M padded to 64 (capacity buffers): 64-row tiles, zero tail waste
-    dim3 grid(N / 128, cdiv(M, 64), E);
+    dim3 grid(N / 128, cdiv(M, 64), E), block(256);
     hipLaunchKernelGGL((batched_gemm_nt_kernel<EPI, 1, 4, 4, 2>), grid, block, 0, stream,
                        (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
                        (const float*)bias, (__bf16*)DER, M, N, K);
