@@ -97,6 +97,22 @@ class ElasticGroup:
                 alive.append(s)
         return alive
 
+    def _stable_alive(self, settle: float = 0.5, max_wait: float = 10.0) -> List[int]:
+        """Debounced membership view: right after a peer dies, survivors'
+        announce traversals stall on the dead DHT node (rpc timeout) long
+        enough for EVERY lease to lapse — a transient window where each rank
+        sees only itself. Require two consecutive identical views before
+        acting on membership."""
+        deadline = time.monotonic() + max_wait
+        prev = self.alive_slots()
+        while time.monotonic() < deadline and not self._stop.is_set():
+            time.sleep(settle)
+            cur = self.alive_slots()
+            if cur == prev and self.slot in cur:
+                return cur
+            prev = cur
+        return sorted(set(prev + [self.slot]))
+
     # ---- generation protocol -------------------------------------------
     @property
     def _gen_key(self) -> str:
@@ -125,9 +141,9 @@ class ElasticGroup:
             dist.destroy_process_group()
             self.group = None
         while time.monotonic() < t_end and not self._stop.is_set():
-            alive = self.alive_slots()
+            self.announce()  # our own lease may have lapsed during a stall
+            alive = self._stable_alive()
             if self.slot not in alive:
-                self.announce()
                 alive = sorted(set(alive + [self.slot]))
             if expected is not None and len(alive) < expected:
                 time.sleep(0.25)

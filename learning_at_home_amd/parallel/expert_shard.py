@@ -197,6 +197,28 @@ class ExpertParallelMoE(nn.Module):
         self.aux_loss: Optional[torch.Tensor] = None
         self.last_counts: Optional[torch.Tensor] = None
         self._capacity_cache: dict = {}
+        # elastic membership (parallel/elastic.py): member_slots = ordered
+        # alive rank slots; None = static full membership. Dead slots'
+        # experts are masked out of routing (reference semantics: dead
+        # experts vanish; they return when the slot rejoins).
+        self.total_slots = self.world_size
+        self.member_slots: Optional[list] = None
+        self.register_buffer("expert_alive", torch.ones(num_experts, dtype=torch.bool),
+                             persistent=False)
+
+    def update_membership(self, group, member_slots: list, total_slots: int) -> None:
+        """Adopt a new generation: process group over `member_slots` (ordered
+        as the group's ranks), expert ownership fixed by slot id."""
+        self.group = group
+        self.total_slots = total_slots
+        self.member_slots = list(member_slots)
+        self.world_size = len(member_slots)
+        self.rank = None  # not needed: a2a uses group rank order
+        per_slot = self.num_experts // total_slots
+        alive = torch.zeros(self.num_experts, dtype=torch.bool)
+        for s in member_slots:
+            alive[s * per_slot : (s + 1) * per_slot] = True
+        self.expert_alive.copy_(alive.to(self.expert_alive.device))
 
     def capacity(self, T: int) -> int:
         """Per-(source rank, expert) capacity, padded to the GEMM tile (128
@@ -217,7 +239,15 @@ class ExpertParallelMoE(nn.Module):
         E, k, W = self.num_experts, self.k, self.world_size
         C = self.capacity(T)
 
+        elastic = self.member_slots is not None
+        members = self.member_slots if elastic else list(range(W))
+        P = self.num_experts // self.total_slots  # experts per slot (fixed)
+
         logits = self.gate(flat).float()
+        if elastic and not bool(self.expert_alive.all()):
+            # dead slots' experts leave the routing (reference: dead experts
+            # vanish until their server re-announces)
+            logits = logits.masked_fill(~self.expert_alive.to(logits.device), -1e9)
         if self.training and self.router_noise_std > 0:
             # exploration noise (Shazeer-style noisy gating): every expert
             # keeps receiving tokens early on, preventing rich-get-richer
@@ -226,28 +256,45 @@ class ExpertParallelMoE(nn.Module):
         topk_idx, topk_w, probs = Fops.gating_topk(logits, k)
         with torch.no_grad():
             pos, counts, inv = Fops.assign_slots(topk_idx, E, C)
+            if elastic:
+                # belt-and-braces: any slot that still picked a dead expert
+                # (e.g. every candidate masked) is treated as dropped
+                dead_sel = ~self.expert_alive.to(pos.device)[topk_idx.long()]
+                pos = pos.masked_fill(dead_sel, -1)
         self.last_counts = counts
 
         compute_dtype = torch.bfloat16 if flat.is_cuda else flat.dtype
         xbuf = Fops.dispatch(flat.to(compute_dtype), topk_idx, pos, inv, k)  # [E, C, D]
 
-        if W > 1:
-            xbuf = _AllToAll.apply(xbuf, self.group)  # [W*E_local, C, D] -> recv
+        if W > 1 or elastic:
+            if elastic and members != list(range(self.total_slots)):
+                xsend = torch.cat([xbuf[s * P : (s + 1) * P] for s in members])
+            else:
+                xsend = xbuf
+            if W > 1:
+                xsend = _AllToAll.apply(xsend, self.group)  # [W*P, C, D] recv
             xrecv = (
-                xbuf.view(W, self.num_local, C, D).permute(1, 0, 2, 3)
-                .reshape(self.num_local, W * C, D).contiguous()
+                xsend.view(W, P, C, D).permute(1, 0, 2, 3)
+                .reshape(P, W * C, D).contiguous()
             )
         else:
             xrecv = xbuf
 
         ybuf_local = Fops.batched_ffn(xrecv, self.experts)
 
-        if W > 1:
+        if W > 1 or elastic:
             ysend = (
-                ybuf_local.view(self.num_local, W, C, D).permute(1, 0, 2, 3)
-                .reshape(E, C, D).contiguous()
+                ybuf_local.view(P, W, C, D).permute(1, 0, 2, 3)
+                .reshape(W * P, C, D).contiguous()
             )
-            ybuf = _AllToAll.apply(ysend, self.group)
+            if W > 1:
+                ysend = _AllToAll.apply(ysend, self.group)
+            if elastic and members != list(range(self.total_slots)):
+                ybuf = torch.zeros(E, C, D, dtype=ysend.dtype, device=ysend.device)
+                for i, s in enumerate(members):
+                    ybuf[s * P : (s + 1) * P] = ysend[i * P : (i + 1) * P]
+            else:
+                ybuf = ysend
         else:
             ybuf = ybuf_local
 

@@ -138,3 +138,45 @@ class TestExpertParallelGloo:
         """2-rank expert-parallel forward == single-process all-local forward
         (token dispatch over all_to_all, experts sharded)."""
         mp.spawn(_ep_worker, args=(2, str(tmp_path)), nprocs=2, join=True)
+
+
+def _elastic_moe_worker(rank, world):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29541")
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    solo_groups = [dist.new_group([r]) for r in range(world)]
+    torch.manual_seed(7)
+    moe = ExpertParallelMoE(d_model=32, ffn_dim=64, num_experts=4, k=2,
+                            capacity_factor=8.0, aux_loss_coef=0.0)
+    moe.eval()
+    x = torch.randn(10, 32)
+    y_full = moe(x)
+    assert torch.isfinite(y_full).all()
+
+    # partition: each rank continues alone; the other slot's experts are dead
+    moe.update_membership(solo_groups[rank], [rank], total_slots=world)
+    y_solo = moe(x)
+    assert torch.isfinite(y_solo).all()
+    # dead experts must receive no tokens
+    per_slot = 4 // world
+    dead = [e for e in range(4) if e // per_slot != rank]
+    for e in dead:
+        assert int(moe.last_counts[e]) == 0, f"dead expert {e} got tokens"
+    # gradients still flow through the surviving experts
+    y2 = moe(x.requires_grad_(True))
+    y2.sum().backward()
+    assert moe.experts.W1.grad is not None
+
+    # full membership restored
+    moe.update_membership(None, [0, 1], total_slots=world)
+    moe.group = None
+    moe.world_size = world
+    moe.group = dist.group.WORLD
+    y_back = moe(x.detach())
+    assert torch.isfinite(y_back).all()
+    assert torch.allclose(y_back, y_full, atol=1e-4), "restored membership differs"
+    dist.destroy_process_group()
+
+
+class TestElasticMoE:
+    def test_membership_shrink_and_restore(self):
+        mp.spawn(_elastic_moe_worker, args=(2,), nprocs=2, join=True)
