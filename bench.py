@@ -178,7 +178,7 @@ def main():
     t0 = time.perf_counter()
     last_loss = None
     for i in range(args.steps):
-        last_loss = train_step(args.warmup + i)
+        last_loss = train_step(args.warmup + i).detach()
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
