@@ -59,21 +59,20 @@ __global__ void dispatch_bwd_kernel(const __bf16* __restrict__ dxbuf,  // [E, C,
   int t = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   if (t >= T) return;
   int lane = threadIdx.x & 63;
-  for (int c = lane * 4; c < D; c += 64 * 4) {
-    float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int c = lane * 8; c < D; c += 64 * 8) {
+    float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
     for (int j = 0; j < k; ++j) {
       int p = pos[(long)t * k + j];
       if (p < 0) continue;
       int e = topk_idx[(long)t * k + j];
-      const __bf16* src = dxbuf + ((long)e * C + p) * D + c;
-      bf16x4 v = *(const bf16x4*)src;
+      bf16x8 v = *(const bf16x8*)(dxbuf + ((long)e * C + p) * D + c);
 #pragma unroll
-      for (int q = 0; q < 4; ++q) acc[q] += bf2f(v[q]);
+      for (int q = 0; q < 8; ++q) acc[q] += bf2f(v[q]);
     }
-    bf16x4 out;
+    bf16x8 out;
 #pragma unroll
-    for (int q = 0; q < 4; ++q) out[q] = f2bf(acc[q]);
-    *(bf16x4*)(dx + (long)t * D + c) = out;
+    for (int q = 0; q < 8; ++q) out[q] = f2bf(acc[q]);
+    *(bf16x8*)(dx + (long)t * D + c) = out;
   }
 }
 
@@ -94,21 +93,21 @@ __global__ void combine_fwd_kernel(const __bf16* __restrict__ ybuf,   // [E, C, 
   // softmax mass, an unclamped 1/s amplifies combine gradients by up to
   // 1/w_min and detonates the router (observed at flagship scale)
   float inv_s = (s > 0.f) ? 1.0f / fmaxf(s, 0.2f) : 0.f;
-  for (int c = lane * 4; c < D; c += 64 * 4) {
-    float acc[4] = {0.f, 0.f, 0.f, 0.f};
+  for (int c = lane * 8; c < D; c += 64 * 8) {
+    float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
     for (int j = 0; j < k; ++j) {
       int p = pos[(long)t * k + j];
       if (p < 0) continue;
       int e = topk_idx[(long)t * k + j];
       float wj = w[(long)t * k + j] * inv_s;
-      bf16x4 v = *(const bf16x4*)(ybuf + ((long)e * C + p) * D + c);
+      bf16x8 v = *(const bf16x8*)(ybuf + ((long)e * C + p) * D + c);
 #pragma unroll
-      for (int q = 0; q < 4; ++q) acc[q] += wj * bf2f(v[q]);
+      for (int q = 0; q < 8; ++q) acc[q] += wj * bf2f(v[q]);
     }
-    bf16x4 out;
+    bf16x8 out;
 #pragma unroll
-    for (int q = 0; q < 4; ++q) out[q] = f2bf(acc[q]);
-    *(bf16x4*)(y + (long)t * D + c) = out;
+    for (int q = 0; q < 8; ++q) out[q] = f2bf(acc[q]);
+    *(bf16x8*)(y + (long)t * D + c) = out;
   }
 }
 
@@ -135,11 +134,11 @@ __global__ void combine_bwd_kernel(const __bf16* __restrict__ dy,     // [T, D]
 
   // <dy, y> once per token
   float dot_dy_y = 0.f;
-  for (int c = lane * 4; c < D; c += 64 * 4) {
-    bf16x4 a = *(const bf16x4*)(dy + (long)t * D + c);
-    bf16x4 b = *(const bf16x4*)(y + (long)t * D + c);
+  for (int c = lane * 8; c < D; c += 64 * 8) {
+    bf16x8 a = *(const bf16x8*)(dy + (long)t * D + c);
+    bf16x8 b = *(const bf16x8*)(y + (long)t * D + c);
 #pragma unroll
-    for (int q = 0; q < 4; ++q) dot_dy_y += bf2f(a[q]) * bf2f(b[q]);
+    for (int q = 0; q < 8; ++q) dot_dy_y += bf2f(a[q]) * bf2f(b[q]);
   }
 #pragma unroll
   for (int off = 32; off > 0; off >>= 1) dot_dy_y += __shfl_xor(dot_dy_y, off);
@@ -154,16 +153,16 @@ __global__ void combine_bwd_kernel(const __bf16* __restrict__ dy,     // [T, D]
     float wj = w[(long)t * k + j] * inv_s;
     long boff = ((long)e * C + p) * D;
     float dot_dy_yj = 0.f;
-    for (int c = lane * 4; c < D; c += 64 * 4) {
-      bf16x4 a = *(const bf16x4*)(dy + (long)t * D + c);
-      bf16x4 b = *(const bf16x4*)(ybuf + boff + c);
-      bf16x4 out;
+    for (int c = lane * 8; c < D; c += 64 * 8) {
+      bf16x8 a = *(const bf16x8*)(dy + (long)t * D + c);
+      bf16x8 b = *(const bf16x8*)(ybuf + boff + c);
+      bf16x8 out;
 #pragma unroll
-      for (int q = 0; q < 4; ++q) {
+      for (int q = 0; q < 8; ++q) {
         dot_dy_yj += bf2f(a[q]) * bf2f(b[q]);
         out[q] = f2bf(wj * bf2f(a[q]));
       }
-      *(bf16x4*)(dybuf + boff + c) = out;
+      *(bf16x8*)(dybuf + boff + c) = out;
     }
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) dot_dy_yj += __shfl_xor(dot_dy_yj, off);

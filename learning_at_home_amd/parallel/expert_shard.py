@@ -72,6 +72,68 @@ class ExpertStack(nn.Module):
         self.main_grad_b2 = torch.zeros_like(self.W2_bias, device=dev)
         self._gpu_state_ready = True
 
+    # ---- per-expert checkpoint interop (server/checkpoints.py format) ---
+    def export_expert_state(self, idx: int) -> dict:
+        """FFNExpert-compatible state_dict of local expert `idx` — the same
+        on-disk layout the socket Server writes, so EP shards and RPC servers
+        can exchange checkpoints (SURVEY.md §5 'byte-compatible')."""
+        self.sync_master_weights()
+        return {
+            "fc1.weight": self.W1[idx].detach().cpu().clone(),
+            "fc1.bias": self.W1_bias[idx].detach().cpu().clone(),
+            "fc2.weight": self.W2[idx].detach().cpu().clone(),
+            "fc2.bias": self.W2_bias[idx].detach().cpu().clone(),
+        }
+
+    @torch.no_grad()
+    def import_expert_state(self, idx: int, sd: dict) -> None:
+        self.W1.data[idx].copy_(sd["fc1.weight"].to(self.W1.device))
+        self.W1_bias.data[idx].copy_(sd["fc1.bias"].to(self.W1_bias.device))
+        self.W2.data[idx].copy_(sd["fc2.weight"].to(self.W2.device))
+        self.W2_bias.data[idx].copy_(sd["fc2.bias"].to(self.W2_bias.device))
+        if self._gpu_state_ready:
+            from ..ops import _C
+
+            self.W1c[idx].copy_(self.W1.data[idx].bfloat16().to(self.W1c.device))
+            self.W2c[idx].copy_(self.W2.data[idx].bfloat16().to(self.W2c.device))
+            self.W1t[idx].copy_(self.W1c[idx].t().contiguous())
+            self.W2t[idx].copy_(self.W2c[idx].t().contiguous())
+            self.b1_f32[idx].copy_(self.W1_bias.data[idx].to(self.b1_f32.device))
+            self.b2_f32[idx].copy_(self.W2_bias.data[idx].to(self.b2_f32.device))
+
+    def save_checkpoints(self, checkpoint_dir, uid_prefix: str, first_global_idx: int = 0,
+                         update_count: int = 0) -> None:
+        import time as _time
+        from pathlib import Path
+
+        ts = _time.strftime("%Y%m%d_%H%M%S")
+        for i in range(self.num_local):
+            uid = f"{uid_prefix}.{first_global_idx + i}"
+            d = Path(checkpoint_dir) / uid
+            d.mkdir(parents=True, exist_ok=True)
+            blob = {"model": self.export_expert_state(i), "optimizer": {},
+                    "update_count": update_count}
+            tmp = d / f".tmp_{ts}.pt"
+            torch.save(blob, tmp)
+            import os as _os
+
+            _os.replace(tmp, d / f"checkpoint_{ts}.pt")
+            import shutil as _shutil
+
+            _shutil.copyfile(d / f"checkpoint_{ts}.pt", d / "checkpoint_last.pt")
+
+    def load_checkpoints(self, checkpoint_dir, uid_prefix: str, first_global_idx: int = 0) -> int:
+        from pathlib import Path
+
+        n = 0
+        for i in range(self.num_local):
+            f = Path(checkpoint_dir) / f"{uid_prefix}.{first_global_idx + i}" / "checkpoint_last.pt"
+            if f.exists():
+                blob = torch.load(f, map_location="cpu", weights_only=False)
+                self.import_expert_state(i, blob["model"])
+                n += 1
+        return n
+
     def sync_master_weights(self) -> None:
         """Pull bf16 compute weights back into the fp32 master Parameters
         (needed before state_dict/checkpoint in stochastic-rounding mode)."""

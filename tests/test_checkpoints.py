@@ -60,3 +60,37 @@ class TestCheckpoints:
     def test_load_missing_is_noop(self, tmp_path):
         b = _make_backend()
         assert load_experts({"ffn.0": b}, tmp_path) == 0
+
+
+class TestExpertStackInterop:
+    def test_stack_checkpoint_loads_into_rpc_server_expert(self, tmp_path):
+        """EP shard checkpoints and socket-server checkpoints share one
+        format: a stack-written file restores an FFNExpert byte-for-byte."""
+        import torch
+        from learning_at_home_amd.parallel import ExpertStack
+
+        torch.manual_seed(5)
+        stack = ExpertStack(2, 32, 64)
+        stack.save_checkpoints(tmp_path, "ffn", first_global_idx=4)
+        assert (tmp_path / "ffn.5" / "checkpoint_last.pt").exists()
+
+        b = _make_backend(uid="ffn.5", d=32)
+        assert load_experts({"ffn.5": b}, tmp_path) == 1
+        x = torch.randn(3, 32)
+        with torch.no_grad():
+            y_module = b.expert(x)
+            ref = torch.nn.functional.gelu(x @ stack.W1[1].t() + stack.W1_bias[1],
+                                           approximate="tanh") @ stack.W2[1].t() + stack.W2_bias[1]
+        assert torch.allclose(y_module, ref, atol=1e-6)
+
+    def test_stack_roundtrip(self, tmp_path):
+        import torch
+        from learning_at_home_amd.parallel import ExpertStack
+
+        torch.manual_seed(6)
+        src = ExpertStack(2, 32, 64)
+        src.save_checkpoints(tmp_path, "rt")
+        dst = ExpertStack(2, 32, 64)
+        assert dst.load_checkpoints(tmp_path, "rt") == 2
+        assert torch.equal(src.W1.detach(), dst.W1.detach())
+        assert torch.equal(src.W2_bias.detach(), dst.W2_bias.detach())
