@@ -116,5 +116,6 @@ class ExpertBackend:
 
     def load_state_dict(self, state: Dict[str, Any]) -> None:
         self.expert.load_state_dict(state["model"])
-        self.optimizer.load_state_dict(state["optimizer"])
+        if state.get("optimizer"):  # an EP-shard checkpoint has no torch-optim state
+            self.optimizer.load_state_dict(state["optimizer"])
         self.update_count = int(state.get("update_count", 0))

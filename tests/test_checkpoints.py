@@ -70,7 +70,7 @@ class TestExpertStackInterop:
         from learning_at_home_amd.parallel import ExpertStack
 
         torch.manual_seed(5)
-        stack = ExpertStack(2, 32, 64)
+        stack = ExpertStack(2, 32, 128)  # ffn_dim matches FFNExpert(32)'s 4*d default
         stack.save_checkpoints(tmp_path, "ffn", first_global_idx=4)
         assert (tmp_path / "ffn.5" / "checkpoint_last.pt").exists()
 
