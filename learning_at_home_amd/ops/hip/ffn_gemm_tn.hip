@@ -11,10 +11,7 @@
 // follow-up (see guide T10).
 #include "common.h"
 
-#define TBI 128
-#define TBJ 128
 #define TBK 64  // reduction (m) tile
-#define TN_THREADS 256
 
 // stage one [TBK rows of m][128 cols] tile TRANSPOSED into LDS image
 // [128][TBK]. XOR swizzle on the m-chunk index spreads both the scalar
@@ -23,25 +20,30 @@
 //   chunk' = (m>>3) ^ (i & 7) ^ ((i >> 3) & 7)
 #define TN_OFF(i, m) ((i) * TBK + (((((m) >> 3) ^ ((i) & 7) ^ (((i) >> 3) & 7)) << 3) | ((m) & 7)))
 
+template <int COLS, int THREADS>
 LAH_DEVINL void stage_tile_t(const __bf16* __restrict__ src, __bf16* lds, int col0,
                              int m0, int ld, int tid) {
+  constexpr int CPR = COLS / 8;  // 16-byte chunks per m-row
 #pragma unroll
-  for (int it = 0; it < (TBK * 128) / (TN_THREADS * 8); ++it) {
-    int idx = it * TN_THREADS + tid;  // 16-byte chunk over [64][128]
-    int m = idx >> 4;
-    int i16 = idx & 15;
-    bf16x8 v = *(const bf16x8*)(src + (long)(m0 + m) * ld + col0 + i16 * 8);
+  for (int it = 0; it < (TBK * COLS) / (THREADS * 8); ++it) {
+    int idx = it * THREADS + tid;  // 16-byte chunk over [TBK][COLS]
+    int m = idx / CPR;
+    int ic = idx % CPR;
+    bf16x8 v = *(const bf16x8*)(src + (long)(m0 + m) * ld + col0 + ic * 8);
 #pragma unroll
-    for (int q = 0; q < 8; ++q) lds[TN_OFF(i16 * 8 + q, m)] = v[q];
+    for (int q = 0; q < 8; ++q) lds[TN_OFF(ic * 8 + q, m)] = v[q];
   }
 }
 
-template <bool ACC>
-__launch_bounds__(TN_THREADS, 2) __global__
+template <bool ACC, int WM, int WN, int FM, int FN>
+__launch_bounds__(WM * WN * 64, (WM * WN > 4) ? 1 : 2) __global__
 void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
                             const __bf16* __restrict__ B,  // [E, M, J]
                             float* __restrict__ C,         // [E, I, J]
                             int M, int I, int J) {
+  constexpr int TBI = WM * FM * 16;
+  constexpr int TBJ = WN * FN * 16;
+  constexpr int THREADS = WM * WN * 64;
   __shared__ __attribute__((aligned(16))) __bf16 smem[2 * (TBI + TBJ) * TBK];
   const int HALF = (TBI + TBJ) * TBK;
 
@@ -51,21 +53,21 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int wr = wave >> 1, wc = wave & 1;
+  const int wr = wave / WN, wc = wave % WN;
 
   const __bf16* Ae = A + (long)e * M * I;
   const __bf16* Be = B + (long)e * M * J;
 
-  f32x4 acc[4][4];
+  f32x4 acc[FM][FN];
 #pragma unroll
-  for (int a = 0; a < 4; ++a)
+  for (int a = 0; a < FM; ++a)
 #pragma unroll
-    for (int b = 0; b < 4; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+    for (int b = 0; b < FN; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
 
   const int nk = M / TBK;
   int cur = 0;
-  stage_tile_t(Ae, smem, i0, 0, I, tid);
-  stage_tile_t(Be, smem + TBI * TBK, j0, 0, J, tid);
+  stage_tile_t<TBI, THREADS>(Ae, smem, i0, 0, I, tid);
+  stage_tile_t<TBJ, THREADS>(Be, smem + TBI * TBK, j0, 0, J, tid);
   __syncthreads();
 
   const int l15 = lane & 15;
@@ -75,27 +77,27 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
     const __bf16* la = smem + cur * HALF;
     const __bf16* lb = smem + cur * HALF + TBI * TBK;
     if (t + 1 < nk) {
-      stage_tile_t(Ae, smem + (cur ^ 1) * HALF, i0, (t + 1) * TBK, I, tid);
-      stage_tile_t(Be, smem + (cur ^ 1) * HALF + TBI * TBK, j0, (t + 1) * TBK, J, tid);
+      stage_tile_t<TBI, THREADS>(Ae, smem + (cur ^ 1) * HALF, i0, (t + 1) * TBK, I, tid);
+      stage_tile_t<TBJ, THREADS>(Be, smem + (cur ^ 1) * HALF + TBI * TBK, j0, (t + 1) * TBK, J, tid);
     }
 #pragma unroll
     for (int kk = 0; kk < TBK / 32; ++kk) {
-      bf16x8 afrag[4], bfrag[4];
+      bf16x8 afrag[FM], bfrag[FN];
       const int km = kk * 32 + lk8;
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
-        int row = wr * 64 + mi * 16 + l15;
+      for (int mi = 0; mi < FM; ++mi) {
+        int row = wr * FM * 16 + mi * 16 + l15;
         afrag[mi] = *(const bf16x8*)(la + TN_OFF(row, km));
       }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
-        int row = wc * 64 + ni * 16 + l15;
+      for (int ni = 0; ni < FN; ++ni) {
+        int row = wc * FN * 16 + ni * 16 + l15;
         bfrag[ni] = *(const bf16x8*)(lb + TN_OFF(row, km));
       }
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < FM; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
+        for (int ni = 0; ni < FN; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[mi], bfrag[ni],
                                                                 acc[mi][ni], 0, 0, 0);
     }
@@ -108,13 +110,13 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
   // 2 barriers cost more than the store coalescing gains)
   float* Ce = C + (long)e * I * J;
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
+  for (int mi = 0; mi < FM; ++mi) {
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
-      int col = j0 + wc * 64 + ni * 16 + l15;
+    for (int ni = 0; ni < FN; ++ni) {
+      int col = j0 + wc * FN * 16 + ni * 16 + l15;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int row = i0 + wr * 64 + mi * 16 + (lane >> 4) * 4 + r;
+        int row = i0 + wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
         long off = (long)row * J + col;
         Ce[off] = ACC ? Ce[off] + acc[mi][ni][r] : acc[mi][ni][r];
       }
@@ -122,13 +124,24 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
   }
 }
 
+template <bool ACC>
+static void launch_tn(const void* A, const void* B, void* C, int E, int M, int I, int J,
+                      hipStream_t stream) {
+  if (I % 256 == 0 && J % 256 == 0) {
+    dim3 grid(J / 256, I / 256, E), block(512);
+    hipLaunchKernelGGL((batched_gemm_tn_kernel<ACC, 2, 4, 8, 4>), grid, block, 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, (float*)C, M, I, J);
+  } else {
+    dim3 grid(J / 128, I / 128, E), block(256);
+    hipLaunchKernelGGL((batched_gemm_tn_kernel<ACC, 2, 2, 4, 4>), grid, block, 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, (float*)C, M, I, J);
+  }
+}
+
 extern "C" void lah_batched_gemm_tn(const void* A, const void* B, void* C, int E, int M,
                                     int I, int J, int accumulate, hipStream_t stream) {
-  dim3 grid(J / TBJ, I / TBI, E), block(TN_THREADS);
   if (accumulate)
-    hipLaunchKernelGGL(batched_gemm_tn_kernel<true>, grid, block, 0, stream,
-                       (const __bf16*)A, (const __bf16*)B, (float*)C, M, I, J);
+    launch_tn<true>(A, B, C, E, M, I, J, stream);
   else
-    hipLaunchKernelGGL(batched_gemm_tn_kernel<false>, grid, block, 0, stream,
-                       (const __bf16*)A, (const __bf16*)B, (float*)C, M, I, J);
+    launch_tn<false>(A, B, C, E, M, I, J, stream);
 }

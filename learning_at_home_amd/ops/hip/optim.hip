@@ -175,16 +175,34 @@ __global__ void transpose_bf16_kernel(const __bf16* __restrict__ in, __bf16* __r
   }
 }
 
-// column sum for bias grads: db[e][c] = sum_m d[e][m][c], fp32 out
+// column sum for bias grads: db[e][c] = sum_m d[e][m][c], fp32 out.
+// Each thread owns 8 consecutive columns (bf16x8 loads, 16 B/lane) and the
+// m-loop is unrolled 4 deep to keep several loads in flight.
 __global__ void colsum_kernel(const __bf16* __restrict__ d, float* __restrict__ db,
                               int M, int C) {
   int e = blockIdx.z;
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  const __bf16* de = d + (long)e * M * C;
-  float acc = 0.f;
-  for (int m = 0; m < M; ++m) acc += bf2f(de[(long)m * C + c]);
-  db[(long)e * C + c] = acc;
+  int c8 = blockIdx.x * blockDim.x + threadIdx.x;  // 8-column chunk index
+  if (c8 * 8 >= C) return;
+  const __bf16* de = d + (long)e * M * C + c8 * 8;
+  float acc[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  int m = 0;
+  for (; m + 4 <= M; m += 4) {
+    bf16x8 v0 = *(const bf16x8*)(de + (long)(m + 0) * C);
+    bf16x8 v1 = *(const bf16x8*)(de + (long)(m + 1) * C);
+    bf16x8 v2 = *(const bf16x8*)(de + (long)(m + 2) * C);
+    bf16x8 v3 = *(const bf16x8*)(de + (long)(m + 3) * C);
+#pragma unroll
+    for (int q = 0; q < 8; ++q)
+      acc[q] += bf2f(v0[q]) + bf2f(v1[q]) + bf2f(v2[q]) + bf2f(v3[q]);
+  }
+  for (; m < M; ++m) {
+    bf16x8 v = *(const bf16x8*)(de + (long)m * C);
+#pragma unroll
+    for (int q = 0; q < 8; ++q) acc[q] += bf2f(v[q]);
+  }
+  float* out = db + (long)e * C + c8 * 8;
+#pragma unroll
+  for (int q = 0; q < 8; q += 4) *(f4*)(out + q) = *(f4*)(acc + q);
 }
 
 extern "C" void lah_sgd_step(void* master, const void* grad, void* momentum,
@@ -223,7 +241,7 @@ extern "C" void lah_transpose_bf16(const void* in, void* out, int E, int R, int 
 
 extern "C" void lah_colsum(const void* d, void* db, int E, int M, int C,
                            hipStream_t stream) {
-  dim3 grid(cdiv(C, 256), 1, E), block(256);
+  dim3 grid(cdiv(C / 8, 256), 1, E), block(256);
   hipLaunchKernelGGL(colsum_kernel, grid, block, 0, stream, (const __bf16*)d, (float*)db,
                      M, C);
 }
