@@ -2,5 +2,6 @@
 
 from .expert import RemoteExpert, rpc_call
 from .moe import RemoteMixtureOfExperts
+from .beam_search import MoEBeamSearcher
 
-__all__ = ["RemoteExpert", "RemoteMixtureOfExperts", "rpc_call"]
+__all__ = ["RemoteExpert", "RemoteMixtureOfExperts", "MoEBeamSearcher", "rpc_call"]

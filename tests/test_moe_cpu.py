@@ -237,3 +237,27 @@ class TestFaultTolerance:
                 s1.dht.shutdown()
             client_dht.shutdown()
             dht.shutdown()
+
+
+class TestBeamSearcher:
+    def test_multidim_grid_beam(self, moe_grid):
+        """2x2 uid grid 'bs.i.j': beam search finds the live best-scoring
+        experts through DHT prefix probes."""
+        import torch
+        from learning_at_home_amd.client import MoEBeamSearcher
+
+        dht, servers, client_dht = moe_grid
+        # declare a 2x2 grid on server 0's DHT (liveness only; endpoints of s1)
+        s = servers[0]
+        uids = ["bs.0.0", "bs.0.1", "bs.1.0"]  # bs.1.1 intentionally missing
+        s.dht.declare_experts(uids, s.addr, s.port, expiration=60)
+        searcher = MoEBeamSearcher(client_dht, "bs", (2, 2), beam_size=4)
+        scores = [torch.tensor([0.1, 2.0]), torch.tensor([3.0, 1.0])]
+        found = searcher.find_best_experts(scores, k_best=3)
+        found_uids = [u for u, ep, sc in found]
+        # best scoring would be bs.1.0 (2.0 + 3.0); bs.1.1 (3.0) is dead
+        assert found_uids[0] == "bs.1.0"
+        assert "bs.1.1" not in found_uids
+        assert len(found) == 3
+        scores_out = [sc for _, _, sc in found]
+        assert scores_out == sorted(scores_out, reverse=True)
