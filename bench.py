@@ -57,7 +57,7 @@ def main():
     p.add_argument("--experts", type=int, default=256)
     p.add_argument("--k", type=int, default=4)
     p.add_argument("--seq", type=int, default=1024)
-    p.add_argument("--batch", type=int, default=64, help="per-GPU batch size")
+    p.add_argument("--batch", type=int, default=96, help="per-GPU batch size")
     p.add_argument("--vocab", type=int, default=32000)
     p.add_argument("--capacity-factor", type=float, default=1.5)
     p.add_argument("--aux-coef", type=float, default=0.01)
