@@ -92,17 +92,23 @@ class TransformerDMoE(nn.Module):
         self.lm_head = nn.Linear(config.d_model, config.vocab_size, bias=False)
         self.lm_head.weight = self.embed.weight  # tied
 
-    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+    def hidden_states(self, input_ids: torch.Tensor) -> torch.Tensor:
         B, S = input_ids.shape
         pos = torch.arange(S, device=input_ids.device)
         x = self.embed(input_ids) + self.pos_embed(pos)[None]
         for block in self.blocks:
             x = block(x)
-        return self.lm_head(self.ln_f(x))
+        return self.ln_f(x)
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        return self.lm_head(self.hidden_states(input_ids))
 
     def loss(self, input_ids: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
-        logits = self.forward(input_ids)
-        loss = F.cross_entropy(logits.flatten(0, 1).float(), labels.flatten())
+        from ..ops.fused_ce import chunked_linear_cross_entropy
+
+        hidden = self.hidden_states(input_ids)
+        # chunked fused linear+CE: never materializes the full [T, V] logits
+        loss = chunked_linear_cross_entropy(hidden, self.lm_head.weight, labels)
         # accumulate load-balance auxiliary losses collected by MoE layers
         for block in self.blocks:
             aux = getattr(block.moe, "aux_loss", None)
