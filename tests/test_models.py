@@ -30,7 +30,7 @@ class TestChunkedCE:
                                     seq_len=16, num_experts=4, k_best=2, ffn_dim=64)
         model = TransformerDMoE(cfg)
         ids = torch.randint(0, 128, (2, 17))
-        loss = model.loss(ids[:, :-1], ids[:, 1:])
+        loss = model.loss(ids[:, :-1], ids[:, 1:], fused_ce=True)
         loss.backward()
         assert torch.isfinite(loss)
         assert model.embed.weight.grad is not None  # tied lm_head got grads
