@@ -22,9 +22,15 @@ __version__ = "0.1.0"
 from .dht import DHT
 from .runtime import ExpertBackend, Runtime, TaskPool
 from .server import Server
-from .client import RemoteExpert, RemoteMixtureOfExperts
+from .client import MoEBeamSearcher, RemoteExpert, RemoteMixtureOfExperts
 from .averaging import DecentralizedAverager
-from .optim import DecentralizedSGD
+from .optim import DecentralizedSGD, MasterSGD
+
+# reference-prototype codename aliases (the prototype called these
+# TesseractServer / TesseractNetwork / TesseractRuntime)
+TesseractServer = Server
+TesseractNetwork = DHT
+from .runtime import Runtime as TesseractRuntime  # noqa: E402
 
 __all__ = [
     "DHT",
@@ -34,6 +40,7 @@ __all__ = [
     "Server",
     "RemoteExpert",
     "RemoteMixtureOfExperts",
+    "MoEBeamSearcher",
     "DecentralizedAverager",
     "DecentralizedSGD",
     "__version__",

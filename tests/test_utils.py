@@ -140,3 +140,25 @@ class TestTimedStorage:
         s.store("c", 3, t0 + 15)
         assert len(s) == 2
         assert s.get("a") == (None, None)  # soonest-to-expire evicted
+
+
+class TestMonitor:
+    def test_stage_timers_cpu(self):
+        from learning_at_home_amd.utils.monitor import StageTimers
+
+        t = StageTimers(use_gpu_events=False)
+        with t("work"):
+            sum(range(1000))
+        with t("work"):
+            pass
+        s = t.summary()
+        assert s["work"]["count"] == 2 and s["work"]["total_s"] >= 0
+
+    def test_expert_load_stats(self):
+        import torch
+        from learning_at_home_amd.utils.monitor import expert_load_stats
+
+        stats = expert_load_stats(torch.tensor([10, 10, 10, 10]), capacity=8)
+        assert stats["mean_load"] == 10.0 and stats["cv"] == 0.0
+        assert stats["dropped_frac"] == 0.2  # 2 over capacity per expert
+        assert stats["entropy_frac"] == 1.0
