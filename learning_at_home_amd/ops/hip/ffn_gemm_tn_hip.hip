@@ -5,11 +5,16 @@
 // written DIRECTLY — no bmm round trip, no bf16->fp32 convert, no add pass.)
 //
 // Both operands are m-major in global memory (the reduction dim is the
-// leading one), so staging transposes each [64_m][128_i] tile into an
-// i-major LDS image [128][64] that the MFMA fragment reads then consume
-// exactly like the NT kernel does. The transposed ds_write_b16 scatter pays
-// LDS write conflicts; acceptable v1 — the ds_read_b64_tr_b16 path is the
-// follow-up (see guide T10).
+// leading one), so staging transposes each [64_m][cols] tile into a
+// cols-major LDS image that the MFMA fragments then consume exactly like
+// the NT kernel does (b128 reads, XOR-swizzled).
+//
+// An alternative ds_read_b64_tr_b16 variant (glds lane-linear staging +
+// hardware-transposed fragment reads; probe-verified semantics) is kept
+// below behind LAH_TN_TR16=1: it is numerically correct but measured
+// SLOWER (481 vs 578 TF at M=1536 — tr-read bank behavior + the opaque
+// asm loads defeating hipcc's LDS wait scheduling), so the ds_write
+// transpose staging stays the default.
 #include "common.h"
 
 #define TBK 64  // reduction (m) tile
@@ -125,9 +130,166 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
   }
 }
 
+// ---------------------------------------------------------------------------
+// tr16 variant: NATURAL (lane-linear) [m][cols] LDS images staged by
+// global_load_lds, fragments delivered TRANSPOSED by ds_read_b64_tr_b16.
+// Probe-verified semantics (experiments/probes/probe_tr16.hip): within each
+// 16-lane group, lane l=(a<<2)|b receives element b of the 4-bf16 runs
+// addressed by lanes a, a+4, a+8, a+12 — so per-lane address
+//   elem(m0 + (g>>2), i0 + 4*(g&3))
+// over a row-major [m][i] image yields frag[j] = A[m0+j][i0+g] for j=0..3.
+// ---------------------------------------------------------------------------
+#define TRN_CHUNK(row, c, CPR) (((row) * (CPR) + ((c) ^ ((row) & 7))) * 8)
+
+template <int COLS, int THREADS>
+LAH_DEVINL void stage_tile_lin(const __bf16* __restrict__ src, __bf16* lds, int col0,
+                               int m0, int ld, int tid) {
+  constexpr int CPR = COLS / 8;  // 16-byte chunks per m-row
+#pragma unroll
+  for (int it = 0; it < (TBK * COLS) / (THREADS * 8); ++it) {
+    int idx = it * THREADS + tid;
+    int row = idx / CPR;
+    int c8 = (idx % CPR) ^ (row & 7);  // source-side swizzle (read applies same XOR)
+    glds16(src + (long)(m0 + row) * ld + col0 + c8 * 8, lds + idx * 8);
+  }
+}
+
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4v;
+
+template <int COLS>
+LAH_DEVINL bf16x8 tr16_frag(const __bf16* lds, int mbase, int i0, int lane) {
+  // per-lane element address for the two 4-row reads (j = 0..3 and 4..7)
+  constexpr int CPR = COLS / 8;
+  const int g = lane & 15;
+  const int icol = i0 + 4 * (g & 3);
+  const int row1 = mbase + (g >> 2);
+  const int row2 = row1 + 4;
+  const int e1 = TRN_CHUNK(row1, icol >> 3, CPR) + (icol & 7);
+  const int e2 = TRN_CHUNK(row2, icol >> 3, CPR) + (icol & 7);
+  unsigned a1 = (unsigned)(unsigned long long)(lds + e1);
+  unsigned a2 = (unsigned)(unsigned long long)(lds + e2);
+  unsigned long long v1, v2;
+  asm volatile("ds_read_b64_tr_b16 %0, %2\n\tds_read_b64_tr_b16 %1, %3"
+               : "=&v"(v1), "=&v"(v2) : "v"(a1), "v"(a2));
+  // caller must s_waitcnt lgkmcnt(0) + sched_barrier before consuming
+  union { unsigned long long u[2]; bf16x8 f; } pack;
+  pack.u[0] = v1;
+  pack.u[1] = v2;
+  return pack.f;
+}
+
+template <bool ACC, int WM, int WN, int FM, int FN>
+__launch_bounds__(WM * WN * 64, (WM * WN > 4) ? 1 : 2) __global__
+void batched_gemm_tn_tr_kernel(const __bf16* __restrict__ A,  // [E, M, I]
+                               const __bf16* __restrict__ B,  // [E, M, J]
+                               float* __restrict__ C,         // [E, I, J]
+                               int M, int I, int J) {
+  constexpr int TBI = WM * FM * 16;
+  constexpr int TBJ = WN * FN * 16;
+  constexpr int THREADS = WM * WN * 64;
+  __shared__ __attribute__((aligned(16))) __bf16 smem[2 * (TBI + TBJ) * TBK];
+  const int HALF = (TBI + TBJ) * TBK;
+
+  const int e = blockIdx.z;
+  const int j0 = blockIdx.x * TBJ;
+  const int i0 = blockIdx.y * TBI;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave / WN, wc = wave % WN;
+
+  const __bf16* Ae = A + (long)e * M * I;
+  const __bf16* Be = B + (long)e * M * J;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int a = 0; a < FM; ++a)
+#pragma unroll
+    for (int b = 0; b < FN; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = M / TBK;
+  int cur = 0;
+  stage_tile_lin<TBI, THREADS>(Ae, smem, i0, 0, I, tid);
+  stage_tile_lin<TBJ, THREADS>(Be, smem + TBI * TBK, j0, 0, J, tid);
+
+  constexpr int GLDS_PER_STAGE = (TBI * TBK) / (THREADS * 8) + (TBJ * TBK) / (THREADS * 8);
+  const int l15 = lane & 15;
+  const int lk8 = (lane >> 4) * 8;  // this lane group's 8-row m slice
+
+  for (int t = 0; t < nk; ++t) {
+    if (t + 1 < nk) {
+      stage_tile_lin<TBI, THREADS>(Ae, smem + (cur ^ 1) * HALF, i0, (t + 1) * TBK, I, tid);
+      stage_tile_lin<TBJ, THREADS>(Be, smem + (cur ^ 1) * HALF + TBI * TBK, j0,
+                                   (t + 1) * TBK, J, tid);
+      if constexpr (GLDS_PER_STAGE == 8)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    asm volatile("" ::: "memory");
+    const __bf16* la = smem + cur * HALF;
+    const __bf16* lb = smem + cur * HALF + TBI * TBK;
+#pragma unroll
+    for (int kk = 0; kk < TBK / 32; ++kk) {
+      bf16x8 afrag[FM], bfrag[FN];
+      const int mbase = kk * 32 + lk8;
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi)
+        afrag[mi] = tr16_frag<TBI>(la, mbase, wr * FM * 16 + mi * 16, lane);
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni)
+        bfrag[ni] = tr16_frag<TBJ>(lb, mbase, wc * FN * 16 + ni * 16, lane);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);  // rule 18: pin MFMAs after the wait
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < FN; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[mi], bfrag[ni],
+                                                                acc[mi][ni], 0, 0, 0);
+    }
+    asm volatile("" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+
+  float* Ce = C + (long)e * I * J;
+#pragma unroll
+  for (int mi = 0; mi < FM; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < FN; ++ni) {
+      int col = j0 + wc * FN * 16 + ni * 16 + l15;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = i0 + wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
+        long off = (long)row * J + col;
+        Ce[off] = ACC ? Ce[off] + acc[mi][ni][r] : acc[mi][ni][r];
+      }
+    }
+  }
+}
+
+static int lah_use_tn_tr16() {
+  static int v = -1;
+  if (v < 0) {
+    const char* env = getenv("LAH_TN_TR16");
+    v = (env != nullptr && env[0] == '1') ? 1 : 0;
+  }
+  return v;
+}
+
 template <bool ACC>
 static void launch_tn(const void* A, const void* B, void* C, int E, int M, int I, int J,
                       hipStream_t stream) {
+  if (lah_use_tn_tr16() && I % 256 == 0 && J % 256 == 0) {
+    dim3 grid(J / 256, I / 256, E), block(512);
+    hipLaunchKernelGGL((batched_gemm_tn_tr_kernel<ACC, 2, 4, 8, 4>), grid, block, 0,
+                       stream, (const __bf16*)A, (const __bf16*)B, (float*)C, M, I, J);
+    return;
+  }
   if (I % 256 == 0 && J % 256 == 0) {
     dim3 grid(J / 256, I / 256, E), block(512);
     hipLaunchKernelGGL((batched_gemm_tn_kernel<ACC, 2, 4, 8, 4>), grid, block, 0, stream,
