@@ -4,11 +4,10 @@ recomputed-free fused backward). The CPU path doubles as the numerics
 reference for the GPU tests (tests/test_ops_gpu.py compares kernel outputs
 against these functions in fp32).
 
-GEMM policy: the four activation-side GEMMs of the expert FFN (fwd x2,
-dh/dx) are the hand-written MFMA NT kernels with fused bias/GELU epilogues
-(ffn_gemm.hip); the weight-gradient GEMMs (dW1/dW2) currently go through
-torch.bmm (hipBLASLt — a plain library GEMM, no fusion) pending the TN
-kernel; bias grads via the colsum kernel.
+GEMM policy: ALL six expert-FFN GEMMs are hand-written MFMA kernels — the
+four activation-side NT GEMMs with fused bias/GELU epilogues (ffn_gemm.hip)
+and the two weight-gradient TN GEMMs writing fp32 grads in place
+(ffn_gemm_tn.hip); bias grads via the vectorized colsum kernel (optim.hip).
 """
 from __future__ import annotations
 

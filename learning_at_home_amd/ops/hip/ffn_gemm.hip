@@ -109,18 +109,26 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
                             __bf16* __restrict__ C,        // [E, M, N]
                             const float* __restrict__ bias,  // [E, N] or null
                             __bf16* __restrict__ DER,      // [E, M, N] store/read
-                            int M, int N, int K) {
+                            int M, int N, int K, int swz) {
   constexpr int BM = WM * FM * 16;
   constexpr int BN = WN * FN * 16;
   constexpr int THREADS = WM * WN * 64;
   __shared__ __attribute__((aligned(16))) __bf16 smem[2 * (BM + BN) * BK];
   const int HALF = (BM + BN) * BK;
 
-  // (XCD-aware remap was A/B-tested here and REGRESSED: the multi-dim grid
-  // dispatch order does not match the 1-D b%8 placement the remap assumes)
-  const int e = blockIdx.z;
-  const int n0 = blockIdx.x * BN;
-  const int m0 = blockIdx.y * BM;
+  // Optional fine-grained XCD remap (LAH_XCD_SWZ=1): within each group of
+  // 64 hardware blocks (dispatch = lin%8 -> XCD, probe-verified), XCD x
+  // takes 8 CONSECUTIVE tiles — neighbor tiles share one L2 while staying
+  // temporally co-resident. (A coarse per-XCD mega-chunk remap regressed.)
+  int lin = (blockIdx.z * gridDim.y + blockIdx.y) * gridDim.x + blockIdx.x;
+  if (swz) {
+    int total = gridDim.x * gridDim.y * gridDim.z;
+    if (lin < (total & ~63))
+      lin = (lin & ~63) | ((lin & 7) << 3) | ((lin >> 3) & 7);
+  }
+  const int e = lin / (gridDim.x * gridDim.y);
+  const int n0 = (lin % gridDim.x) * BN;
+  const int m0 = ((lin / gridDim.x) % gridDim.y) * BM;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -228,6 +236,15 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
 }
 
 
+static int lah_xcd_swz() {
+  static int v = -1;
+  if (v < 0) {
+    const char* env = getenv("LAH_XCD_SWZ");
+    v = (env != nullptr && env[0] == '1') ? 1 : 0;
+  }
+  return v;
+}
+
 // runtime switch for the big-tile variant (A/B tuning; see bench_kernels.py)
 static int lah_use_256_tile() {
   static int v = -1;
@@ -246,17 +263,17 @@ static void launch_nt(const void* A, const void* B, void* C, const void* bias,
     dim3 grid(N / 256, M / 256, E), block(512);
     hipLaunchKernelGGL((batched_gemm_nt_kernel<EPI, 2, 4, 8, 4>), grid, block, 0, stream,
                        (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
-                       (const float*)bias, (__bf16*)DER, M, N, K);
+                       (const float*)bias, (__bf16*)DER, M, N, K, lah_xcd_swz());
   } else if (M % 128 == 0) {
     dim3 grid(N / 128, M / 128, E), block(256);
     hipLaunchKernelGGL((batched_gemm_nt_kernel<EPI, 2, 2, 4, 4>), grid, block, 0, stream,
                        (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
-                       (const float*)bias, (__bf16*)DER, M, N, K);
+                       (const float*)bias, (__bf16*)DER, M, N, K, lah_xcd_swz());
   } else {  // M padded to 64 (capacity buffers): 64-row tiles, zero tail waste
     dim3 grid(N / 128, cdiv(M, 64), E), block(256);
     hipLaunchKernelGGL((batched_gemm_nt_kernel<EPI, 1, 4, 4, 2>), grid, block, 0, stream,
                        (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
-                       (const float*)bias, (__bf16*)DER, M, N, K);
+                       (const float*)bias, (__bf16*)DER, M, N, K, lah_xcd_swz());
   }
 }
 

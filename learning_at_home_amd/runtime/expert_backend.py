@@ -8,10 +8,10 @@ Key reference semantics preserved exactly:
 - after every backward batch the expert's own optimizer steps immediately
   (asynchronous / delayed-gradient SGD — no global barrier anywhere).
 
-MI355X path: when the wrapped module is a `models.FFNExpertHIP` (or any
-module exposing `.fused_forward/.fused_backward`), compute runs through the
-hand-written CDNA4 grouped-FFN kernels; otherwise plain PyTorch ops (CPU
-tests, config #1 in BASELINE.json).
+MI355X note: this RPC-shaped backend serves the socket path (cross-host /
+CPU parity, config #1) with plain PyTorch modules; the intra-node hot path
+holds experts STACKED in parallel/ExpertStack and runs the hand-written
+CDNA4 grouped-FFN kernels (ops/hip) — see parallel/expert_shard.py.
 """
 from __future__ import annotations
 

@@ -57,7 +57,14 @@ class ConnectionHandler:
 
     def shutdown(self) -> None:
         if self._loop is not None and self._loop.is_running():
-            self._loop.call_soon_threadsafe(self._loop.stop)
+            def _stop() -> None:
+                if self._server is not None:
+                    self._server.close()
+                for task in asyncio.all_tasks(self._loop):
+                    task.cancel()
+                self._loop.call_soon(self._loop.stop)
+
+            self._loop.call_soon_threadsafe(_stop)
         if self._thread is not None:
             self._thread.join(timeout=5.0)
         self._waiters.shutdown(wait=False)

@@ -1,0 +1,20 @@
+"""The driver depends on bench.py's JSON contract — lock it with a test."""
+import json
+import subprocess
+import sys
+
+
+def test_bench_json_contract():
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=300,
+    )
+    line = out.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    for key in ("metric", "value", "unit", "n_gpus", "steps", "warmup", "ms_per_step",
+                "higher_is_better", "scaling", "vs_baseline", "dtype", "data", "config"):
+        assert key in d, f"missing key {key}"
+    assert d["metric"].startswith("training tokens/sec, Transformer-DMoE 256-expert")
+    assert d["higher_is_better"] is True and d["scaling"] == "weak"
+    assert d["value"] > 0 and d["steps"] == 2 and d["warmup"] == 1
+    assert "global_batch" in d["config"] and "seq_len" in d["config"]
