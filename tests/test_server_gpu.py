@@ -35,3 +35,19 @@ def test_server_on_gpu_roundtrip():
             server.dht.shutdown()
         client_dht.shutdown()
         dht.shutdown()
+
+
+def test_expert_stack_checkpoint_roundtrip_gpu(tmp_path):
+    """GPU ExpertStack (with compute copies) -> checkpoint -> fresh GPU stack:
+    bf16/transposed copies refreshed on import."""
+    from learning_at_home_amd.parallel import ExpertStack
+
+    torch.manual_seed(3)
+    src = ExpertStack(2, 128, 256).to("cuda:0")
+    src.init_gpu_state(master_weights=False)  # SR mode: masters on CPU
+    src.save_checkpoints(tmp_path, "g")
+    dst = ExpertStack(2, 128, 256).to("cuda:0")
+    dst.init_gpu_state(master_weights=True)
+    assert dst.load_checkpoints(tmp_path, "g") == 2
+    assert torch.allclose(dst.W1c.float().cpu(), src.W1c.float().cpu(), atol=1e-2)
+    assert torch.equal(dst.W1t.cpu(), dst.W1c.transpose(1, 2).contiguous().cpu())
