@@ -78,7 +78,8 @@ class DHT(mp.Process):
         return state
 
     # ---- lifecycle ----------------------------------------------------
-    def run_in_background(self, await_ready: bool = True, timeout: float = 15.0) -> None:
+    def run_in_background(self, await_ready: bool = True, timeout: float = 120.0) -> None:
+        # generous: a spawn-context child re-imports torch (~2-10 s, worse under load)
         self.start()
         if await_ready and not self.ready.wait(timeout):
             raise TimeoutError("DHT process failed to start")
