@@ -57,7 +57,7 @@ LAH_DEVINL void gelu_both(float x, float* g, float* dg) {
 }
 
 // coalesced bf16x8 store pass: ERB rows of the fp32 LDS tile -> global
-template <int EPI, int BN, int THREADS>
+template <int EPI, int BN, int CSTR, int THREADS>
 LAH_DEVINL void emit_rows(const float* ctile, __bf16* Ce, __bf16* De, const float* be,
                           int row0, int n0, int M, int N, int nrows, int tid, int /*unused*/) {
   constexpr int TPR = BN / 16;  // threads per row (16 cols x bf16x8 pair each)
@@ -80,7 +80,7 @@ LAH_DEVINL void emit_rows(const float* ctile, __bf16* Ce, __bf16* De, const floa
       if (EPI == EPI_GELU_GRAD) dv = *(const bf16x8*)(De + goff + h * 8);
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
-        float v = ctile[rr * BN + ecol + h * 8 + q];
+        float v = ctile[rr * CSTR + ecol + h * 8 + q];
         if (EPI == EPI_BIAS) {
           outv[h][q] = f2bf(v + bv[h * 8 + q]);
         } else if (EPI == EPI_BIAS_GELU) {
@@ -206,8 +206,13 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
   // staging LDS (BM*BN*4 <= 2*(BM+BN)*BK*2 for both instantiated shapes).
   // rows per epilogue pass: limited by the fp32 LDS image (multi-pass for
   // the 256x256 tile whose full image would need 256 KiB)
+  // fp32 LDS image with a +4-float row pad: an unpadded BN-stride put every
+  // store-pass lane group on 2 banks (16-way; PMC: 100M SQ_LDS_BANK_CONFLICT
+  // on the fwd1 epilogue); stride BN+4 spreads rows by one bank.
+  constexpr int CSTR = BN + 4;
   constexpr int LDS_BYTES = 2 * (BM + BN) * BK * 2;
-  constexpr int ERB = (LDS_BYTES / (BN * 4)) < BM ? (LDS_BYTES / (BN * 4)) : BM;
+  constexpr int ERB0 = LDS_BYTES / (CSTR * 4);
+  constexpr int ERB = (ERB0 >= BM) ? BM : (ERB0 >= 128 ? 128 : (ERB0 >= 64 ? 64 : 32));
   static_assert(ERB >= 16 && BM % ERB == 0, "bad epilogue row split");
   float* ctile = (float*)smem;
   const int ncol = (tid & 7) * 16;
@@ -222,7 +227,7 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int row = wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
-          if (row / ERB == ep) ctile[(row % ERB) * BN + col] = acc[mi][ni][r];
+          if (row / ERB == ep) ctile[(row % ERB) * CSTR + col] = acc[mi][ni][r];
         }
       }
     }
@@ -232,7 +237,7 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
     __bf16* De = (EPI == EPI_BIAS_GELU || EPI == EPI_GELU_GRAD)
                      ? DER + (long)e * M * N : nullptr;
     const float* be = (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) ? bias + (long)e * N : nullptr;
-    emit_rows<EPI, BN, THREADS>(ctile, Ce, De, be, m0 + ep * ERB, n0, M, N, ERB, tid, ncol);
+    emit_rows<EPI, BN, CSTR, THREADS>(ctile, Ce, De, be, m0 + ep * ERB, n0, M, N, ERB, tid, ncol);
   }
 }
 

@@ -43,7 +43,7 @@ __global__ void sgd_step_kernel(float* __restrict__ master, const float* __restr
 __global__ void sgd_step_t_kernel(float* __restrict__ master, const float* __restrict__ grad,
                                   __bf16* __restrict__ out, __bf16* __restrict__ outT,
                                   int R, int C, float lr, float weight_decay) {
-  __shared__ __bf16 tile[64][64 + 8];
+  __shared__ __bf16 tile[64][64 + 2];
   int e = blockIdx.z;
   long base = (long)e * R * C;
   int r0 = blockIdx.y * 64, c0 = blockIdx.x * 64;
@@ -104,7 +104,7 @@ LAH_DEVINL __bf16 round_sr(float v, unsigned rnd16) {
 __global__ void sgd_step_sr_t_kernel(__bf16* __restrict__ param, const float* __restrict__ grad,
                                      __bf16* __restrict__ paramT, int R, int C, float lr,
                                      float weight_decay, unsigned seed) {
-  __shared__ __bf16 tile[64][64 + 8];
+  __shared__ __bf16 tile[64][64 + 2];
   int e = blockIdx.z;
   long base = (long)e * R * C;
   int r0 = blockIdx.y * 64, c0 = blockIdx.x * 64;
@@ -149,7 +149,7 @@ __global__ void sgd_step_sr_t_kernel(__bf16* __restrict__ param, const float* __
 #define TP 64
 __global__ void transpose_bf16_kernel(const __bf16* __restrict__ in, __bf16* __restrict__ out,
                                       int R, int C) {
-  __shared__ __bf16 tile[TP][TP + 8];
+  __shared__ __bf16 tile[TP][TP + 2];
   int e = blockIdx.z;
   int r0 = blockIdx.y * TP, c0 = blockIdx.x * TP;
   const __bf16* ine = in + (long)e * R * C;
