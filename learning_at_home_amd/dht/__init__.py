@@ -194,16 +194,17 @@ class DHT(mp.Process):
     async def _rpc_declare_experts(node: DHTNode, uids, addr, port, expiration) -> bool:
         expiration_time = get_dht_time() + expiration
         endpoint = [addr, port]
-        tasks = []
+        items = []
         seen_prefixes = set()
         for uid in uids:
-            tasks.append(node.store(EXPERT_PREFIX + uid, endpoint, expiration_time))
+            items.append((EXPERT_PREFIX + uid, endpoint, expiration_time))
             for prefix in uid_prefixes(uid):
                 if prefix not in seen_prefixes:
                     seen_prefixes.add(prefix)
-                    tasks.append(node.store(PREFIX_PREFIX + prefix, endpoint, expiration_time))
-        results = await asyncio.gather(*tasks)
-        return all(results)
+                    items.append((PREFIX_PREFIX + prefix, endpoint, expiration_time))
+        # batched multi-key store: one traversal per key only on neighbor-cache
+        # miss, one store RPC per destination peer (not per key x replica)
+        return await node.store_many(items, cache_neighbors=True)
 
     @staticmethod
     async def _rpc_get_experts(node: DHTNode, uids):

@@ -25,6 +25,12 @@ class DHTNode:
         self.port: int = None
         self.alpha = 3
         self.k = 20
+        self.replication = 8  # peers a value is stored on (<= k)
+        # neighbor cache for periodic re-announces: key_id -> (expiry, peers).
+        # A server re-declaring hundreds of expert uids every announce period
+        # would otherwise pay a full iterative traverse per key per period.
+        self._store_neighbors: dict = {}
+        self._store_neighbors_ttl = 60.0
 
     @classmethod
     async def create(
@@ -144,17 +150,74 @@ class DHTNode:
         return nearest
 
     # ------------------------------------------------------------------
-    async def store(self, key: Any, value: Any, expiration_time: float) -> bool:
-        """Store (key → value, expiring) on the k nearest peers (incl. self)."""
+    async def store(self, key: Any, value: Any, expiration_time: float,
+                    cache_neighbors: bool = False) -> bool:
+        """Store (key → value, expiring) on the `replication` nearest peers
+        (incl. self). With cache_neighbors=True the traversal result is
+        reused for `_store_neighbors_ttl` seconds — the re-announce fast path
+        (liveness leases re-store the same keys every announce period)."""
+        import time as _time
+
         key_id = DHTID.generate(key)
-        nearest = await self.find_nearest_nodes(key_id)
+        nearest = None
+        if cache_neighbors:
+            hit = self._store_neighbors.get(int(key_id))
+            if hit is not None and hit[0] > _time.monotonic():
+                nearest = hit[1]
+        if nearest is None:
+            nearest = await self.find_nearest_nodes(key_id, k=self.replication)
+            if cache_neighbors:
+                self._store_neighbors[int(key_id)] = (
+                    _time.monotonic() + self._store_neighbors_ttl, nearest
+                )
         stored_locally = self.protocol.storage.store(key_id, value, expiration_time)
         if not nearest:
             return stored_locally
         results = await asyncio.gather(
             *(self.protocol.call_store(ep, key_id, value, expiration_time) for _, ep in nearest)
         )
-        return any(bool(r) for r in results) or stored_locally
+        ok = any(bool(r) for r in results) or stored_locally
+        if cache_neighbors and not ok:
+            self._store_neighbors.pop(int(key_id), None)  # stale set: re-traverse next time
+        return ok
+
+    async def store_many(self, items, cache_neighbors: bool = True) -> bool:
+        """Store many (key, value, expiration_time) records efficiently: one
+        traversal per key only on cache miss, then ONE batched store RPC per
+        destination peer instead of one RPC per (key, replica). This is the
+        announce path for servers declaring hundreds of expert uids."""
+        import time as _time
+
+        targets: Dict[Any, list] = {}  # endpoint -> [[key_b, value, exp], ...]
+        ok_local = True
+        for key, value, expiration_time in items:
+            key_id = DHTID.generate(key)
+            nearest = None
+            if cache_neighbors:
+                hit = self._store_neighbors.get(int(key_id))
+                if hit is not None and hit[0] > _time.monotonic():
+                    nearest = hit[1]
+            if nearest is None:
+                nearest = await self.find_nearest_nodes(key_id, k=self.replication)
+                if cache_neighbors:
+                    self._store_neighbors[int(key_id)] = (
+                        _time.monotonic() + self._store_neighbors_ttl, nearest
+                    )
+            ok_local &= self.protocol.storage.store(key_id, value, expiration_time)
+            for _, ep in nearest:
+                targets.setdefault(tuple(ep), []).append(
+                    [key_id.to_bytes(), value, expiration_time]
+                )
+
+        async def send_batches(ep, batch):
+            results = []
+            for s0 in range(0, len(batch), 64):  # keep datagrams small
+                res = await self.protocol.call_store_many(ep, batch[s0 : s0 + 64])
+                results.append(res is not None and all(res))
+            return all(results)
+
+        replies = await asyncio.gather(*(send_batches(ep, b) for ep, b in targets.items()))
+        return (any(replies) if replies else False) or ok_local
 
     async def get(self, key: Any, sufficient_expiration: Optional[float] = None) -> Tuple[Optional[Any], Optional[float]]:
         """Iterative find_value; returns (value, expiration) or (None, None).

@@ -99,6 +99,11 @@ class DHTProtocol(asyncio.DatagramProtocol):
         res = await self.call(endpoint, "store", key_id.to_bytes(), value, expiration)
         return None if res is None else bool(res)
 
+    async def call_store_many(self, endpoint, items) -> Optional[list]:
+        """items: list of [key_id_bytes, value, expiration]; one datagram."""
+        res = await self.call(endpoint, "store_many", items)
+        return None if res is None else list(res)
+
     async def call_find_node(self, endpoint, key_id: DHTID) -> Optional[List[Tuple[int, Any]]]:
         res = await self.call(endpoint, "find_node", key_id.to_bytes())
         if res is None:
@@ -127,6 +132,11 @@ class DHTProtocol(asyncio.DatagramProtocol):
         elif rpc == "store":
             key_b, val, expiration = args
             value = self.storage.store(DHTID.from_bytes(key_b), val, float(expiration))
+        elif rpc == "store_many":
+            value = [
+                self.storage.store(DHTID.from_bytes(kb), val, float(exp))
+                for kb, val, exp in args[0]
+            ]
         elif rpc == "find_node":
             key = DHTID.from_bytes(args[0])
             nearest = self.routing_table.get_nearest_neighbors(int(key), self.bucket_size, exclude=(int(sender),))
