@@ -72,6 +72,10 @@ def main():
                         "stochastic-rounding SGD, the fast path)")
     p.add_argument("--metrics", action="store_true",
                    help="print per-stage timers + expert-load stats to stderr")
+    p.add_argument("--no-dht", action="store_true",
+                   help="skip the per-rank Kademlia DHT peer + expert announce "
+                        "(north-star parity: each GPU rank hosts a shard AND a "
+                        "DHT peer with liveness-leased expert uids)")
     args = p.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -99,6 +103,44 @@ def main():
             num_experts=8, k_best=2, ffn_dim=256,
         )
         B = 2
+
+    # per-rank DHT peer: announce this rank's expert shard under liveness
+    # leases (discovery/elasticity control plane; fail-soft for the bench)
+    dht = None
+    if not args.no_dht:
+        try:
+            from learning_at_home_amd import DHT
+            import threading
+
+            if world_size > 1:
+                seed_ep = [None]
+                if rank == 0:
+                    seed = DHT(start=True)
+                    seed_ep = [list(seed.endpoint)]
+                dist.broadcast_object_list(seed_ep, src=0)
+                peers = [tuple(seed_ep[0])] if rank != 0 else []
+                dht = seed if rank == 0 else DHT(initial_peers=peers, start=True)
+            else:
+                dht = DHT(start=True)
+            n_local = cfg.num_experts // world_size
+            uids = [f"bench_expert.{rank * n_local + i}.{layer}"
+                    for i in range(n_local) for layer in range(cfg.n_layers)]
+            dht.declare_experts(uids, "127.0.0.1", 40000 + rank, expiration=60.0)
+
+            def announce_loop():
+                import time as _t
+
+                while True:
+                    _t.sleep(20.0)
+                    try:
+                        dht.declare_experts(uids, "127.0.0.1", 40000 + rank, expiration=60.0)
+                    except Exception:  # noqa: BLE001
+                        return
+
+            threading.Thread(target=announce_loop, daemon=True).start()
+        except Exception as e:  # noqa: BLE001
+            print(f"[bench] DHT control plane disabled: {e!r}")
+            dht = None
 
     torch.manual_seed(1234 + rank)
     model = build_model(cfg, args, device)
@@ -226,6 +268,8 @@ def main():
             print("[metrics] expert load (layer 0):",
                   json.dumps(expert_load_stats(counts.cpu(), cap)), file=sys.stderr)
 
+    if dht is not None:
+        dht.shutdown()
     if world_size > 1:
         dist.destroy_process_group()
 
