@@ -162,3 +162,47 @@ class TestMonitor:
         assert stats["mean_load"] == 10.0 and stats["cv"] == 0.0
         assert stats["dropped_frac"] == 0.2  # 2 over capacity per expert
         assert stats["entropy_frac"] == 1.0
+
+
+class TestConnectionFraming:
+    def test_sync_roundtrip(self):
+        import socket
+        import threading
+        from learning_at_home_amd.utils import recv_message, send_message
+
+        server = socket.socket()
+        server.bind(("127.0.0.1", 0))
+        server.listen(1)
+        port = server.getsockname()[1]
+        got = {}
+
+        def serve():
+            conn, _ = server.accept()
+            got["msg"] = recv_message(conn)
+            send_message(conn, "pong", b"\x00" * 100000)  # multi-chunk reply
+            conn.close()
+
+        t = threading.Thread(target=serve)
+        t.start()
+        client = socket.create_connection(("127.0.0.1", port))
+        send_message(client, "ping", b"payload")
+        kind, payload = recv_message(client)
+        t.join(5)
+        assert got["msg"] == ("ping", b"payload")
+        assert kind == "pong" and len(payload) == 100000
+        client.close()
+        server.close()
+
+    def test_serializer_pickle_fallback(self):
+        # complex(...) is not msgpack-encodable -> exercises the torch.save path
+        from learning_at_home_amd.utils import PytorchSerializer
+
+        obj = {"z": complex(1, 2)}
+        assert PytorchSerializer.loads(PytorchSerializer.dumps(obj)) == obj
+
+    def test_malformed_dht_datagram_ignored(self):
+        from learning_at_home_amd.dht.protocol import DHTProtocol
+        from learning_at_home_amd.dht.routing import DHTID
+
+        proto = DHTProtocol(DHTID.generate())
+        proto.datagram_received(b"\xff\x00garbage", ("127.0.0.1", 1))  # must not raise
