@@ -46,6 +46,15 @@ def build_model(cfg: TransformerDMoEConfig, args, device, group=None):
 
 
 def main():
+    # DHT peers are mp.Processes; fork after HIP init is unsafe on ROCm, and
+    # set_device precedes DHT creation in the multi-GPU path -> use spawn
+    import multiprocessing as _mp
+
+    try:
+        _mp.set_start_method("spawn", force=True)
+    except RuntimeError:
+        pass
+
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=20)
