@@ -43,17 +43,6 @@ LAH_DEVINL void glds16(const void* gsrc, void* lds_dst) {
       (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
 }
 
-// XCD-aware block remap (guide T1): the dispatcher places linear block b on
-// XCD b%8; remap so each XCD gets a CONTIGUOUS chunk of the tile space
-// (neighboring tiles share operand panels -> L2 hits). Bijective for any n.
-LAH_DEVINL int xcd_swizzle(int lin, int n) {
-  const int NX = 8;
-  int q = n / NX, r = n % NX;
-  int xcd = lin % NX, idx = lin / NX;
-  int base = (xcd < r) ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q;
-  return base + idx;  // bijective: XCD x receives exactly its chunk length
-}
-
 #define HIP_CHECK(expr)                                                              \
   do {                                                                               \
     hipError_t _e = (expr);                                                          \

@@ -58,7 +58,7 @@ LAH_DEVINL void gelu_both(float x, float* g, float* dg) {
 // coalesced bf16x8 store pass: ERB rows of the fp32 LDS tile -> global
 template <int EPI, int BN, int CSTR, int THREADS>
 LAH_DEVINL void emit_rows(const float* ctile, __bf16* Ce, __bf16* De, const float* be,
-                          int row0, int n0, int M, int N, int nrows, int tid, int /*unused*/) {
+                          int row0, int n0, int M, int N, int nrows, int tid) {
   constexpr int TPR = BN / 16;  // threads per row (16 cols x bf16x8 pair each)
   const int ecol = (tid % TPR) * 16;
   // bias is row-invariant: load this thread's 16-wide chunk ONCE
@@ -214,7 +214,6 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
   constexpr int ERB = (ERB0 >= BM) ? BM : (ERB0 >= 128 ? 128 : (ERB0 >= 64 ? 64 : 32));
   static_assert(ERB >= 16 && BM % ERB == 0, "bad epilogue row split");
   float* ctile = (float*)smem;
-  const int ncol = (tid & 7) * 16;
 #pragma unroll
   for (int ep = 0; ep < BM / ERB; ++ep) {
     __syncthreads();  // staging/previous-pass reads done; smem is free
@@ -236,7 +235,7 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
     __bf16* De = (EPI == EPI_BIAS_GELU || EPI == EPI_GELU_GRAD)
                      ? DER + (long)e * M * N : nullptr;
     const float* be = (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) ? bias + (long)e * N : nullptr;
-    emit_rows<EPI, BN, CSTR, THREADS>(ctile, Ce, De, be, m0 + ep * ERB, n0, M, N, ERB, tid, ncol);
+    emit_rows<EPI, BN, CSTR, THREADS>(ctile, Ce, De, be, m0 + ep * ERB, n0, M, N, ERB, tid);
   }
 }
 

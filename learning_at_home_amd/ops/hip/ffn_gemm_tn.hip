@@ -153,8 +153,6 @@ LAH_DEVINL void stage_tile_lin(const __bf16* __restrict__ src, __bf16* lds, int 
   }
 }
 
-typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4v;
-
 template <int COLS>
 LAH_DEVINL bf16x8 tr16_frag(const __bf16* lds, int mbase, int i0, int lane) {
   // per-lane element address for the two 4-row reads (j = 0..3 and 4..7)
