@@ -18,3 +18,20 @@ def test_bench_json_contract():
     assert d["higher_is_better"] is True and d["scaling"] == "weak"
     assert d["value"] > 0 and d["steps"] == 2 and d["warmup"] == 1
     assert "global_batch" in d["config"] and "seq_len" in d["config"]
+
+
+def test_bench_distributed_2proc_gloo():
+    """The driver launches bench via torch.distributed.run for N>1 — lock
+    that path end-to-end (EP all-to-all + averager + DHT control plane)."""
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29561", "bench.py", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=600,
+    )
+    lines = [ln for ln in out.stdout.strip().splitlines() if ln.startswith("{")]
+    assert lines, f"no JSON line; stderr tail: {out.stderr[-500:]}"
+    d = json.loads(lines[-1])
+    assert d["n_gpus"] == 2
+    assert d["value"] > 0
+    assert "ep2" in d["config"]["parallelism"]
