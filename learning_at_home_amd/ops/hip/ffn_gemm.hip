@@ -249,6 +249,182 @@ static int lah_xcd_swz() {
   return v;
 }
 
+// ---------------------------------------------------------------------------
+// 8-phase 256x256 schedule (guide §5 template, reconstructed; env-gated
+// LAH_GEMM_8P=1, OFF by default). MEASURED: numerically correct (3x race
+// screens) but 940 vs 1064 TF at 4k^3 against the 1-phase counted-vmcnt
+// kernel — the guide's own caveat reproduced: a coarse phase split without
+// the fine per-phase ds_read || global_load || MFMA instruction interleave
+// HURTS (-7..-27%). Kept as the starting point for the fine-interleave work:
+// 4 phases per K-tile, one output QUADRANT per phase per wave (16 MFMA),
+// one HALF-tile staged per phase (2 glds/thread), counted vmcnt so 2+ halves
+// stay in flight across RAW barriers. "Halves" are row-bands grouped by
+// consumption: A-half h = rows {h*64+wr*128}, B-half h = cols {h*32+wc*64},
+// so phase p=(rh,ch) consumes exactly A-half rh and B-half ch and each slot
+// frees mid-tile. Issue order per tile: [Ah0, Bh0, Ah1, Bh1] (of tile t+1);
+// waits: vmcnt(4) before phases 0 and 2 (the two oldest halves landed).
+// ---------------------------------------------------------------------------
+template <int EPI>
+__launch_bounds__(512, 1) __global__
+void batched_gemm_nt_8p_kernel(const __bf16* __restrict__ A,  // [E, M, K]
+                               const __bf16* __restrict__ B,  // [E, N, K]
+                               __bf16* __restrict__ C,        // [E, M, N]
+                               const float* __restrict__ bias,
+                               __bf16* __restrict__ DER,
+                               int M, int N, int K) {
+  constexpr int BM = 256, BN = 256, THREADS = 512;
+  constexpr int WM = 2, WN = 4, FM = 8, FN = 4;
+  __shared__ __attribute__((aligned(16))) __bf16 smem[2 * (BM + BN) * BK];
+  const int HALF = (BM + BN) * BK;
+
+  const int e = blockIdx.z;
+  const int n0 = blockIdx.x * BN;
+  const int m0 = blockIdx.y * BM;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave / WN, wc = wave % WN;
+
+  const __bf16* Ae = A + (long)e * M * K;
+  const __bf16* Be = B + (long)e * N * K;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int l15 = lane & 15;
+  const int lk8 = (lane >> 4) * 8;
+  const int nk = K / BK;
+
+  // stage one A-half (rows {h*64 + band*128}, 128 rows) of K-tile t
+  auto stage_a_half = [&](int buf, int t, int h) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int idx = i * THREADS + tid;           // 16B chunk in the half
+      int rl = idx >> 3;                     // 0..127
+      int row = (rl & 63) + ((rl >> 6) << 7) + h * 64;  // actual tile row
+      int c8 = (idx & 7) ^ (row & 7);
+      glds16(Ae + (long)min(m0 + row, M - 1) * K + t * BK + c8 * 8,
+             smem + buf * HALF + row * BK + (idx & 7) * 8);
+    }
+  };
+  // stage one B-half (cols {h*32 + wc*64}, 4 bands of 32 rows)
+  auto stage_b_half = [&](int buf, int t, int h) {
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int idx = i * THREADS + tid;
+      int rl = idx >> 3;                     // 0..127
+      int row = (rl & 31) + ((rl >> 5) << 6) + h * 32;
+      int c8 = (idx & 7) ^ (row & 7);
+      glds16(Be + (long)(n0 + row) * K + t * BK + c8 * 8,
+             smem + buf * HALF + BM * BK + row * BK + (idx & 7) * 8);
+    }
+  };
+
+  // prologue: all 4 halves of tile 0, then (if any) halves Ah0,Bh0 of tile 1
+  stage_a_half(0, 0, 0);
+  stage_b_half(0, 0, 0);
+  stage_a_half(0, 0, 1);
+  stage_b_half(0, 0, 1);
+
+  bf16x8 afrag[FM], bfrag[FN];  // current quadrant row/col-half fragments
+
+  for (int t = 0; t < nk; ++t) {
+    const __bf16* la = smem + (t & 1) * HALF;
+    const __bf16* lb = smem + (t & 1) * HALF + BM * BK;
+    const int nb = (t + 1) & 1;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      // GRAY quadrant order (0,0),(0,1),(1,1),(1,0): each consumed half was
+      // issued >= 3 phases earlier under issue order [Ah0, Bh0, Bh1, Ah1]
+      const int rh = p >> 1, ch = (p & 1) ^ (p >> 1);
+      if (t + 1 < nk) {
+        if (p == 0) stage_a_half(nb, t + 1, 0);
+        else if (p == 1) stage_b_half(nb, t + 1, 0);
+        else if (p == 2) stage_b_half(nb, t + 1, 1);
+        else stage_a_half(nb, t + 1, 1);
+      }
+      // counted waits (per-thread glds ledger, 2 per half): p0 needs the 2
+      // oldest of 4 in-flight halves (Ah0,Bh0), p1 needs Bh1 (all but the 3
+      // newest), p2 needs Ah1 (ditto), p3 consumes already-retired halves
+      if (t + 1 < nk) {
+        if (p == 0) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else if (p == 1 || p == 2) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      } else if (p == 0) {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("" ::: "memory");
+      // quadrant fragments: rows wr*128 + rh*64 (+mi*16), cols wc*64 + ch*32
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        const int kchunk = (kk * 32 + lk8) >> 3;
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi) {
+          int row = wr * 128 + rh * 64 + mi * 16 + l15;
+          afrag[kk * 4 + mi] = *(const bf16x8*)(la + LDS_CHUNK(row, kchunk));
+        }
+#pragma unroll
+        for (int ni = 0; ni < 2; ++ni) {
+          int row = wc * 64 + ch * 32 + ni * 16 + l15;
+          bfrag[kk * 2 + ni] = *(const bf16x8*)(lb + LDS_CHUNK(row, kchunk));
+        }
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+          for (int ni = 0; ni < 2; ++ni)
+            acc[rh * 4 + mi][ch * 2 + ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag[kk * 4 + mi], bfrag[kk * 2 + ni], acc[rh * 4 + mi][ch * 2 + ni],
+                0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      asm volatile("" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  // epilogue identical to the main kernel (fp32 LDS image, padded)
+  constexpr int CSTR = BN + 4;
+  constexpr int ERB = 64;
+  float* ctile = (float*)smem;
+#pragma unroll
+  for (int ep = 0; ep < BM / ERB; ++ep) {
+    __syncthreads();
+#pragma unroll
+    for (int mi = 0; mi < FM; ++mi) {
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni) {
+        int col = wc * FN * 16 + ni * 16 + l15;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int row = wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
+          if (row / ERB == ep) ctile[(row % ERB) * CSTR + col] = acc[mi][ni][r];
+        }
+      }
+    }
+    __syncthreads();
+    __bf16* Ce = C + (long)e * M * N;
+    __bf16* De = (EPI == EPI_BIAS_GELU || EPI == EPI_GELU_GRAD)
+                     ? DER + (long)e * M * N : nullptr;
+    const float* be = (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) ? bias + (long)e * N : nullptr;
+    emit_rows<EPI, BN, CSTR, THREADS>(ctile, Ce, De, be, m0 + ep * ERB, n0, M, N, ERB, tid);
+  }
+}
+
+static int lah_gemm_8p() {
+  static int v = -1;
+  if (v < 0) {
+    const char* env = getenv("LAH_GEMM_8P");
+    v = (env != nullptr && env[0] == '1') ? 1 : 0;
+  }
+  return v;
+}
+
 // runtime switch for the big-tile variant (A/B tuning; see bench_kernels.py)
 static int lah_use_256_tile() {
   static int v = -1;
@@ -262,7 +438,12 @@ static int lah_use_256_tile() {
 template <int EPI>
 static void launch_nt(const void* A, const void* B, void* C, const void* bias,
                       void* DER, int E, int M, int N, int K, hipStream_t stream) {
-  if (M % 256 == 0 && N % 256 == 0 && lah_use_256_tile()) {
+  if (M % 256 == 0 && N % 256 == 0 && lah_gemm_8p()) {
+    dim3 grid(N / 256, M / 256, E), block(512);
+    hipLaunchKernelGGL((batched_gemm_nt_8p_kernel<EPI>), grid, block, 0, stream,
+                       (const __bf16*)A, (const __bf16*)B, (__bf16*)C,
+                       (const float*)bias, (__bf16*)DER, M, N, K);
+  } else if (M % 256 == 0 && N % 256 == 0 && lah_use_256_tile()) {
     // 256x256 tile, 8 waves, 128 KiB LDS, 1 block/CU (guide: high-tile regime)
     dim3 grid(N / 256, M / 256, E), block(512);
     hipLaunchKernelGGL((batched_gemm_nt_kernel<EPI, 2, 4, 8, 4>), grid, block, 0, stream,
