@@ -252,10 +252,12 @@ static int lah_xcd_swz() {
 // ---------------------------------------------------------------------------
 // 8-phase 256x256 schedule (guide §5 template, reconstructed; env-gated
 // LAH_GEMM_8P=1, OFF by default). MEASURED: numerically correct (3x race
-// screens) but 940 vs 1064 TF at 4k^3 against the 1-phase counted-vmcnt
-// kernel — the guide's own caveat reproduced: a coarse phase split without
-// the fine per-phase ds_read || global_load || MFMA instruction interleave
-// HURTS (-7..-27%). Kept as the starting point for the fine-interleave work:
+// screens x2 variants) but 940 TF (reads-after-barrier) / 882 TF (guide
+// order: reads+stage pre-barrier, end-of-phase certification waits) vs
+// 1064 TF for the 1-phase counted-vmcnt kernel at 4k^3 — the guide's own
+// caveat reproduced both ways: a phase split without the exact per-phase
+// ds_read || global_load || MFMA instruction interleave HURTS (-7..-27%).
+// Kept as the starting point for the fine-interleave follow-up:
 // 4 phases per K-tile, one output QUADRANT per phase per wave (16 MFMA),
 // one HALF-tile staged per phase (2 glds/thread), counted vmcnt so 2+ halves
 // stay in flight across RAW barriers. "Halves" are row-bands grouped by
@@ -323,11 +325,14 @@ void batched_gemm_nt_8p_kernel(const __bf16* __restrict__ A,  // [E, M, K]
     }
   };
 
-  // prologue: all 4 halves of tile 0, then (if any) halves Ah0,Bh0 of tile 1
+  // prologue: tile 0's halves in issue order; certify Ah0+Bh0 then barrier
   stage_a_half(0, 0, 0);
   stage_b_half(0, 0, 0);
-  stage_a_half(0, 0, 1);
   stage_b_half(0, 0, 1);
+  stage_a_half(0, 0, 1);
+  if (nk > 1) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
 
   bf16x8 afrag[FM], bfrag[FN];  // current quadrant row/col-half fragments
 
@@ -346,17 +351,11 @@ void batched_gemm_nt_8p_kernel(const __bf16* __restrict__ A,  // [E, M, K]
         else if (p == 2) stage_b_half(nb, t + 1, 1);
         else stage_a_half(nb, t + 1, 1);
       }
-      // counted waits (per-thread glds ledger, 2 per half): p0 needs the 2
-      // oldest of 4 in-flight halves (Ah0,Bh0), p1 needs Bh1 (all but the 3
-      // newest), p2 needs Ah1 (ditto), p3 consumes already-retired halves
-      if (t + 1 < nk) {
-        if (p == 0) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
-        else if (p == 1 || p == 2) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
-      } else if (p == 0) {
-        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-      }
-      __builtin_amdgcn_s_barrier();
-      asm volatile("" ::: "memory");
+      // (the wait certifying THIS phase's halves sat before the PREVIOUS
+      // phase's closing barrier — see the phase tail below — so the
+      // pre-barrier ds_reads here are safe across waves)
+      // guide phase order: ds_reads issued BEFORE the mid barrier (latency
+      // drains during barrier arrival skew), lgkmcnt(0) after it
       // quadrant fragments: rows wr*128 + rh*64 (+mi*16), cols wc*64 + ch*32
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk) {
@@ -372,6 +371,9 @@ void batched_gemm_nt_8p_kernel(const __bf16* __restrict__ A,  // [E, M, K]
           bfrag[kk * 2 + ni] = *(const bf16x8*)(lb + LDS_CHUNK(row, kchunk));
         }
       }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int kk = 0; kk < 2; ++kk)
@@ -383,6 +385,17 @@ void batched_gemm_nt_8p_kernel(const __bf16* __restrict__ A,  // [E, M, K]
                 afrag[kk * 4 + mi], bfrag[kk * 2 + ni], acc[rh * 4 + mi][ch * 2 + ni],
                 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
+      // certify the NEXT phase's halves before the closing barrier: with the
+      // per-tile issue order [Ah0,Bh0,Bh1,Ah1], the half first needed by
+      // phase p+1 is always "all but the 2 newest in flight" => vmcnt(4)
+      // (ends of p0, p1 and p3; nothing new needed entering p3). Tail tiles
+      // have no fresh issues covering the count: drain fully.
+      if (p != 2) {
+        if (t + 1 < nk)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
       asm volatile("" ::: "memory");
       __builtin_amdgcn_s_barrier();
     }
