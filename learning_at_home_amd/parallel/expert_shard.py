@@ -365,8 +365,8 @@ class ExpertParallelMoE(nn.Module):
         aux = None
         if self.aux_loss_coef > 0:
             f = (counts.float() / max(1, T * k)).detach()
-            P = probs.float().mean(dim=0)
-            aux = self.aux_loss_coef * E * torch.dot(f.to(P.device), P)
+            mean_probs = probs.float().mean(dim=0)
+            aux = self.aux_loss_coef * E * torch.dot(f.to(mean_probs.device), mean_probs)
         if self.z_loss_coef > 0:
             # router z-loss (ST-MoE): keeps logit magnitudes bounded so the
             # gating softmax never saturates into a zero-gradient regime
