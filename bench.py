@@ -81,6 +81,9 @@ def main():
                         "stochastic-rounding SGD, the fast path)")
     p.add_argument("--metrics", action="store_true",
                    help="print per-stage timers + expert-load stats to stderr")
+    p.add_argument("--loss-every", type=int, default=0,
+                   help="if >0, rank 0 prints the step loss to stderr every N "
+                        "timed steps (long-horizon stability monitoring)")
     p.add_argument("--no-dht", action="store_true",
                    help="skip the per-rank Kademlia DHT peer + expert announce "
                         "(north-star parity: each GPU rank hosts a shard AND a "
@@ -230,6 +233,10 @@ def main():
     last_loss = None
     for i in range(args.steps):
         last_loss = train_step(args.warmup + i).detach()
+        if args.loss_every and rank == 0 and (i % args.loss_every == 0 or i == args.steps - 1):
+            import sys as _sys
+
+            print(f"[loss] step {i}: {float(last_loss):.4f}", file=_sys.stderr)
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
