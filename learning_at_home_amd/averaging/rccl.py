@@ -12,7 +12,7 @@ contiguous collective (per-link bound, no bucket fragmentation needed below
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 import torch
 import torch.distributed as dist

@@ -10,7 +10,7 @@ the local-beam + get_experts path in moe.py is cheaper.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Sequence, Tuple
+from typing import List, Sequence, Tuple
 
 import torch
 
