@@ -239,6 +239,46 @@ class _BatchedFFNHip(torch.autograd.Function):
         return dxbuf, None
 
 
+@torch.no_grad()
+def fused_ffn_infer(x: torch.Tensor, fc1_weight: torch.Tensor, fc1_bias: torch.Tensor,
+                    fc2_weight: torch.Tensor, fc2_bias: torch.Tensor) -> torch.Tensor:
+    """Inference-only fused FFN for ONE expert (the serving hot path:
+    ExpertBackend.forward on a GPU server): y = gelu(x@W1^T + b1)@W2^T + b2
+    as two MFMA NT launches with fused bias / bias+GELU epilogues — no
+    transposed weight copies needed (nn.Linear stores W[out, in], which is
+    exactly the NT layout). Requires bf16 CUDA tensors with
+    d_model % 128 == 0 and ffn_dim % 128 == 0 (callers gate on
+    `can_fuse_ffn_infer`); arbitrary token count M (64-row tail tiles)."""
+    orig_shape = x.shape
+    d = orig_shape[-1]
+    ffn = fc1_weight.shape[0]
+    x2 = x.reshape(1, -1, d).contiguous()
+    M = x2.shape[1]
+    # EPI_BIAS_GELU also emits gelu'(pre) for training backward; at inference
+    # it is scratch (adding a no-derivative epilogue is not worth a variant)
+    der = torch.empty(1, M, ffn, dtype=torch.bfloat16, device=x.device)
+    h = _C.batched_gemm_nt(x2, fc1_weight.reshape(1, ffn, d).contiguous(),
+                           fc1_bias.reshape(1, ffn).float().contiguous(), der, EPI_BIAS_GELU)
+    y = _C.batched_gemm_nt(h, fc2_weight.reshape(1, d, ffn).contiguous(),
+                           fc2_bias.reshape(1, d).float().contiguous(), None, EPI_BIAS)
+    return y.reshape(orig_shape)
+
+
+def can_fuse_ffn_infer(x: torch.Tensor, expert) -> bool:
+    """True iff `fused_ffn_infer` applies: a gelu FFNExpert, everything bf16
+    on GPU, dims aligned for the NT kernels (N%128, K%64 on both GEMMs)."""
+    if not (HAS_HIP and x.is_cuda and x.dtype == torch.bfloat16 and x.numel() > 0):
+        return False
+    fc1 = getattr(expert, "fc1", None)
+    fc2 = getattr(expert, "fc2", None)
+    if fc1 is None or fc2 is None or getattr(expert, "activation", None) != "gelu":
+        return False
+    d, ffn = fc1.weight.shape[1], fc1.weight.shape[0]
+    return (fc1.weight.dtype == torch.bfloat16 and fc2.weight.dtype == torch.bfloat16
+            and fc1.bias is not None and fc2.bias is not None
+            and d % 128 == 0 and ffn % 128 == 0 and x.shape[-1] == d)
+
+
 def batched_ffn(xbuf: torch.Tensor, stack) -> torch.Tensor:
     """xbuf [E, C, D] -> y [E, C, D] through each expert's FFN."""
     if xbuf.is_cuda and HAS_HIP:

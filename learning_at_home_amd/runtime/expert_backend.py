@@ -60,8 +60,20 @@ class ExpertBackend:
 
     # ------------------------------------------------------------------
     def forward(self, *inputs: torch.Tensor) -> Tuple[torch.Tensor, ...]:
-        """Inference-mode forward. No state is kept (backward recomputes)."""
+        """Inference-mode forward. No state is kept (backward recomputes).
+
+        Serving hot path: a gelu FFNExpert hosted bf16 on a GPU runs through
+        the fused MFMA NT kernels (ops.functional.fused_ffn_infer) — two
+        launches, fused bias/GELU — instead of four torch ops; backward
+        recompute stays on autograd (it needs the graph anyway)."""
         with torch.no_grad():
+            if len(inputs) == 1:
+                from ..ops.functional import can_fuse_ffn_infer, fused_ffn_infer
+
+                if can_fuse_ffn_infer(inputs[0], self.expert):
+                    e = self.expert
+                    return (fused_ffn_infer(inputs[0], e.fc1.weight, e.fc1.bias,
+                                            e.fc2.weight, e.fc2.bias),)
             outputs = self.expert(*inputs)
         return outputs if isinstance(outputs, tuple) else (outputs,)
 

@@ -51,3 +51,38 @@ def test_expert_stack_checkpoint_roundtrip_gpu(tmp_path):
     assert dst.load_checkpoints(tmp_path, "g") == 2
     assert torch.allclose(dst.W1c.float().cpu(), src.W1c.float().cpu(), atol=1e-2)
     assert torch.equal(dst.W1t.cpu(), dst.W1c.transpose(1, 2).contiguous().cpu())
+
+
+def test_fused_ffn_infer_matches_reference():
+    """Serving hot path: ExpertBackend.forward of a bf16 gelu FFNExpert on GPU
+    must take the fused MFMA route and match the fp32 torch reference."""
+    from learning_at_home_amd.ops.functional import can_fuse_ffn_infer, fused_ffn_infer
+    from learning_at_home_amd.runtime import ExpertBackend
+    from learning_at_home_amd.utils import BatchTensorDescriptor
+
+    torch.manual_seed(0)
+    d, ffn = 256, 512
+    expert = FFNExpert(d, ffn).to("cuda:0", torch.bfloat16)
+    for M in (1, 4, 37, 64, 129):  # ragged token counts exercise the 64-row tail tiles
+        x = torch.randn(M, d, device="cuda:0", dtype=torch.bfloat16)
+        assert can_fuse_ffn_infer(x, expert)
+        y = fused_ffn_infer(x, expert.fc1.weight, expert.fc1.bias,
+                            expert.fc2.weight, expert.fc2.bias)
+        with torch.no_grad():
+            ref = FFNExpert(d, ffn).to("cuda:0")
+            ref.load_state_dict({k: v.float() for k, v in expert.state_dict().items()})
+            y_ref = ref(x.float())
+        assert y.shape == (M, d)
+        assert torch.allclose(y.float(), y_ref, atol=2e-2, rtol=2e-2), \
+            f"M={M}: max err {(y.float() - y_ref).abs().max().item()}"
+    # the backend routes through the fused path (same numbers, tuple shape)
+    backend = ExpertBackend("fused", expert, torch.optim.SGD(expert.parameters(), lr=0.0),
+                            args_schema=(BatchTensorDescriptor.make(d, dtype="bfloat16"),))
+    x = torch.randn(37, d, device="cuda:0", dtype=torch.bfloat16)
+    (y_backend,) = backend.forward(x)
+    y_direct = fused_ffn_infer(x, expert.fc1.weight, expert.fc1.bias,
+                               expert.fc2.weight, expert.fc2.bias)
+    assert torch.equal(y_backend, y_direct)
+    # misaligned dims fall back cleanly
+    odd = FFNExpert(96, 200).to("cuda:0", torch.bfloat16)
+    assert not can_fuse_ffn_infer(torch.randn(4, 96, device="cuda:0", dtype=torch.bfloat16), odd)
