@@ -32,6 +32,9 @@ def main():
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--k-best", type=int, default=2)
     p.add_argument("--backward", action="store_true")
+    p.add_argument("--device", default=None, help="host experts on this device (e.g. cuda:0)")
+    p.add_argument("--dtype", default=None, choices=["float32", "bfloat16"],
+                   help="host experts in this dtype (bfloat16 + GPU = fused MFMA serving)")
     args = p.parse_args()
 
     num_experts = args.servers * args.experts_per_server
@@ -42,6 +45,8 @@ def main():
             args.experts_per_server, "ffn", args.hidden, expert_prefix="bench",
             expert_offset=s * args.experts_per_server,
             initial_peers=[seed_dht.endpoint], announce_period=5.0, start=True,
+            device=torch.device(args.device) if args.device else None,
+            dtype=getattr(torch, args.dtype) if args.dtype else None,
         ))
 
     results = [0] * args.trainers

@@ -44,14 +44,19 @@ class ExpertBackend:
         self.args_schema = tuple(args_schema)
         self.grad_clip = grad_clip
         self.update_count = 0
+        # hosting dtype: clients always speak their own dtype (usually fp32);
+        # inputs are cast to the expert's parameter dtype at this boundary and
+        # outputs/grads cast back, so a server may host experts in bf16 (the
+        # GPU serving fast path) transparently to every client
+        first_param = next(iter(expert.parameters()), None)
+        self._param_dtype = first_param.dtype if first_param is not None else None
 
         if outputs_schema is None:
             with torch.no_grad():
                 dummy = [schema.make_batch(1) for schema in self.args_schema]
                 device = next(expert.parameters()).device if any(True for _ in expert.parameters()) else torch.device("cpu")
                 dummy = [d.to(device) for d in dummy]
-                out = expert(*dummy)
-                out = out if isinstance(out, tuple) else (out,)
+                out = self.forward(*dummy)  # same boundary casts clients will see
                 outputs_schema = tuple(BatchTensorDescriptor.from_tensor(o) for o in out)
         self.outputs_schema = tuple(outputs_schema)
 
@@ -66,24 +71,40 @@ class ExpertBackend:
         the fused MFMA NT kernels (ops.functional.fused_ffn_infer) — two
         launches, fused bias/GELU — instead of four torch ops; backward
         recompute stays on autograd (it needs the graph anyway)."""
+        client_dtype = inputs[0].dtype if inputs and inputs[0].is_floating_point() else None
         with torch.no_grad():
+            inputs = tuple(self._to_host_dtype(x) for x in inputs)
             if len(inputs) == 1:
                 from ..ops.functional import can_fuse_ffn_infer, fused_ffn_infer
 
                 if can_fuse_ffn_infer(inputs[0], self.expert):
                     e = self.expert
-                    return (fused_ffn_infer(inputs[0], e.fc1.weight, e.fc1.bias,
-                                            e.fc2.weight, e.fc2.bias),)
+                    y = fused_ffn_infer(inputs[0], e.fc1.weight, e.fc1.bias,
+                                        e.fc2.weight, e.fc2.bias)
+                    return (self._to_client_dtype(y, client_dtype),)
             outputs = self.expert(*inputs)
-        return outputs if isinstance(outputs, tuple) else (outputs,)
+        outputs = outputs if isinstance(outputs, tuple) else (outputs,)
+        return tuple(self._to_client_dtype(o, client_dtype) for o in outputs)
+
+    def _to_host_dtype(self, x: torch.Tensor) -> torch.Tensor:
+        if (self._param_dtype is not None and x.is_floating_point()
+                and self._param_dtype.is_floating_point and x.dtype != self._param_dtype):
+            return x.to(self._param_dtype)
+        return x
+
+    @staticmethod
+    def _to_client_dtype(x: torch.Tensor, dtype) -> torch.Tensor:
+        return x.to(dtype) if dtype is not None and x.is_floating_point() and x.dtype != dtype else x
 
     def backward(self, *inputs_and_grads: torch.Tensor) -> Tuple[torch.Tensor, ...]:
         """Recompute forward under enable_grad, backprop the provided output
         grads, step this expert's optimizer, and return input grads."""
         num_inputs = len(self.args_schema)
         inputs, grad_outputs = inputs_and_grads[:num_inputs], inputs_and_grads[num_inputs:]
+        client_dtypes = [x.dtype if x.is_floating_point() else None for x in inputs]
         with torch.enable_grad():
-            detached = [x.detach().requires_grad_(x.dtype.is_floating_point) for x in inputs]
+            detached = [self._to_host_dtype(x.detach()).requires_grad_(x.dtype.is_floating_point)
+                        for x in inputs]
             outputs = self.expert(*detached)
             outputs = outputs if isinstance(outputs, tuple) else (outputs,)
             assert len(outputs) == len(grad_outputs), (
@@ -94,7 +115,8 @@ class ExpertBackend:
                 retain_graph=False, create_graph=False,
             )
         grad_inputs = tuple(
-            x.grad if x.grad is not None else torch.zeros_like(x) for x in detached
+            self._to_client_dtype(x.grad if x.grad is not None else torch.zeros_like(x), dt)
+            for x, dt in zip(detached, client_dtypes)
         )
         self.apply_gradients()
         return grad_inputs

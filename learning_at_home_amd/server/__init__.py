@@ -123,6 +123,7 @@ class Server:
         optim_kwargs: Optional[dict] = None,
         max_batch_size: int = 16384,
         device: Optional[torch.device] = None,
+        dtype: Optional[torch.dtype] = None,
         listen_on: Tuple[str, int] = ("127.0.0.1", 0),
         initial_peers: Sequence[Tuple[str, int]] = (),
         dht: Optional[DHT] = None,
@@ -146,8 +147,11 @@ class Server:
         backends = {}
         for uid in expert_uids:
             expert = block_cls(hidden_dim, **expert_kwargs)
-            if device is not None:
-                expert.to(device)
+            if device is not None or dtype is not None:
+                # dtype=bfloat16 + GPU enables the fused MFMA serving forward
+                # (ExpertBackend casts at the RPC boundary, so fp32 clients
+                # interoperate transparently)
+                expert.to(device=device, dtype=dtype)
             optimizer = optim_cls(expert.parameters(), **optim_kwargs)
             backends[uid] = ExpertBackend(
                 uid, expert, optimizer,

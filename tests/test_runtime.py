@@ -111,6 +111,45 @@ class TestRuntimeLoop:
             futures = [b.forward_pool.submit_task(torch.randn(2, 8)) for _ in range(5)]
             outs = [f.result(timeout=5) for f in futures]
             assert all(o[0].shape == (2, 8) for o in outs)
+            # samples_processed is a monitoring counter incremented AFTER the
+            # batch's futures resolve — eventually consistent, so poll briefly
+            deadline = time.monotonic() + 2.0
+            while rt.samples_processed < 10 and time.monotonic() < deadline:
+                time.sleep(0.01)
             assert rt.samples_processed == 10
         finally:
             rt.shutdown()
+
+
+class TestExpertBackendDtypeBoundary:
+    def test_bf16_hosted_expert_fp32_client(self):
+        """A server may host experts in bf16 (the GPU serving fast path);
+        fp32 clients must see fp32 outputs and fp32 input grads, with values
+        matching an fp32 twin within bf16 tolerance."""
+        import copy
+
+        torch.manual_seed(0)
+        d = 64
+        expert32 = FFNExpert(d, 128)
+        expert16 = copy.deepcopy(expert32).to(torch.bfloat16)
+        opt = torch.optim.SGD(expert16.parameters(), lr=0.0)
+        backend = ExpertBackend(
+            "bf16", expert16, opt,
+            args_schema=(BatchTensorDescriptor.make(d),),
+        )
+        # outputs_schema reflects the CLIENT-side dtype (fp32), not the host dtype
+        assert backend.outputs_schema[0].torch_dtype == torch.float32
+
+        x = torch.randn(5, d)
+        (y,) = backend.forward(x)
+        assert y.dtype == torch.float32
+        with torch.no_grad():
+            y_ref = expert32(x)
+        assert torch.allclose(y, y_ref, atol=0.05, rtol=0.05)
+
+        dy = torch.randn_like(y_ref)
+        (gx,) = backend.backward(x, dy)
+        assert gx.dtype == torch.float32
+        x_ref = x.detach().requires_grad_(True)
+        expert32(x_ref).backward(dy)
+        assert torch.allclose(gx, x_ref.grad, atol=0.05, rtol=0.05)

@@ -86,3 +86,36 @@ def test_fused_ffn_infer_matches_reference():
     # misaligned dims fall back cleanly
     odd = FFNExpert(96, 200).to("cuda:0", torch.bfloat16)
     assert not can_fuse_ffn_infer(torch.randn(4, 96, device="cuda:0", dtype=torch.bfloat16), odd)
+
+
+def test_bf16_server_fp32_client_roundtrip():
+    """End-to-end serving: bf16 GPU-hosted experts (fused MFMA forward path)
+    behind the RPC socket, fp32 client — dtype casts at the backend boundary."""
+    dht = DHT(start=True)
+    server = Server.create(1, "ffn", 256, expert_prefix="bf16srv",
+                           initial_peers=[dht.endpoint],
+                           device=torch.device("cuda:0"), dtype=torch.bfloat16,
+                           announce_period=2.0, start=True)
+    client_dht = DHT(initial_peers=[dht.endpoint], start=True)
+    try:
+        (ep,) = client_dht.get_experts(["bf16srv.0"])
+        assert ep is not None
+        remote = RemoteExpert("bf16srv.0", *ep)
+        x = torch.randn(37, 256, requires_grad=True)  # ragged M -> tail tiles
+        y = remote(x)
+        assert y.dtype == torch.float32 and y.shape == (37, 256)
+        backend = server.expert_backends["bf16srv.0"]
+        with torch.no_grad():
+            ref = FFNExpert(256).to("cuda:0")
+            ref.load_state_dict({k: v.float() for k, v in backend.expert.state_dict().items()})
+            y_ref = ref(x.detach().to("cuda:0")).cpu()
+        assert torch.allclose(y.detach(), y_ref, atol=0.05, rtol=0.05)
+        y.sum().backward()
+        assert x.grad is not None and x.grad.dtype == torch.float32
+        assert torch.isfinite(x.grad).all()
+    finally:
+        server.shutdown()
+        if server.dht is not None:
+            server.dht.shutdown()
+        client_dht.shutdown()
+        dht.shutdown()
