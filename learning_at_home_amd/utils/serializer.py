@@ -39,7 +39,8 @@ def _pack_tensor(t: torch.Tensor) -> bytes:
     t = t.contiguous()
     if t.dtype in _RAW_VIEW_DTYPES:
         dtype_name = _RAW_VIEW_DTYPES[t.dtype]
-        raw = t.view(torch.uint8).numpy().tobytes()
+        # reshape(-1) first: .view(uint8) rejects 0-dim tensors
+        raw = t.reshape(-1).view(torch.uint8).numpy().tobytes()
     else:
         dtype_name = str(t.dtype).replace("torch.", "")
         raw = t.numpy().tobytes()
@@ -50,7 +51,12 @@ def _unpack_tensor(data: bytes) -> torch.Tensor:
     dtype_name, shape, requires_grad, raw = msgpack.unpackb(data, raw=False)
     if dtype_name in _RAW_VIEW_NAMES:
         torch_dtype = _RAW_VIEW_NAMES[dtype_name]
-        flat = torch.frombuffer(bytearray(raw), dtype=torch.uint8).view(torch_dtype)
+        if len(raw) == 0:  # frombuffer rejects empty buffers
+            flat = torch.empty(0, dtype=torch_dtype)
+        else:
+            flat = torch.frombuffer(bytearray(raw), dtype=torch.uint8).view(torch_dtype)
+    elif len(raw) == 0:
+        flat = torch.empty(0, dtype=getattr(torch, dtype_name))
     else:
         np_arr = np.frombuffer(raw, dtype=np.dtype(dtype_name)).copy()
         flat = torch.from_numpy(np_arr)
