@@ -206,3 +206,32 @@ class TestConnectionFraming:
 
         proto = DHTProtocol(DHTID.generate())
         proto.datagram_received(b"\xff\x00garbage", ("127.0.0.1", 1))  # must not raise
+
+
+class TestMonitor:
+    def test_stage_timers_cpu(self):
+        import time as _time
+
+        from learning_at_home_amd.utils.monitor import StageTimers
+
+        timers = StageTimers(use_gpu_events=False)
+        for _ in range(3):
+            with timers("work"):
+                _time.sleep(0.01)
+        timers.flush()
+        s = timers.summary()
+        assert s["work"]["count"] == 3
+        assert s["work"]["total_s"] >= 0.03
+        assert "work" in timers.report()
+
+    def test_expert_load_stats_uniform_and_skewed(self):
+        from learning_at_home_amd.utils.monitor import expert_load_stats
+
+        uniform = expert_load_stats(torch.full((8,), 100.0), capacity=128)
+        assert uniform["dropped_frac"] == 0.0
+        assert uniform["cv"] == 0.0
+        assert abs(uniform["entropy_frac"] - 1.0) < 1e-6
+        skewed = expert_load_stats(torch.tensor([800.0] + [0.0] * 7), capacity=128)
+        assert skewed["max_load"] == 800 and skewed["min_load"] == 0
+        assert skewed["dropped_frac"] == round((800 - 128) / 800, 4)
+        assert skewed["entropy_frac"] < 0.01  # all mass on one expert
