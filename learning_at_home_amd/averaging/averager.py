@@ -231,6 +231,12 @@ class DecentralizedAverager:
                 round_state = self._rounds[group_id] = _RoundState(expected_ranks=-1)
         round_state.add_part(rank, part, weight)
         if not round_state.ready.wait(self.request_timeout + self.averaging_expiration):
+            # drop the placeholder so aborted rounds don't accumulate forever
+            # (idempotent: a concurrent local _run_allreduce holds its own
+            # reference and pops again in its finally block)
+            with self._rounds_lock:
+                if self._rounds.get(group_id) is round_state and round_state.result is None:
+                    self._rounds.pop(group_id, None)
             raise TimeoutError("averaging round timed out waiting for parts")
         return "result", PytorchSerializer.dumps(round_state.result)
 
