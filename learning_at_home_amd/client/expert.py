@@ -48,9 +48,11 @@ class _SocketPool:
     def request(self, endpoint: Endpoint, kind: str, payload: bytes,
                 timeout: float = _DEFAULT_TIMEOUT) -> bytes:
         q = self._get_queue(endpoint)
+        pooled = True
         try:
             sock = q.get_nowait()
         except queue.Empty:
+            pooled = False
             sock = socket.create_connection(endpoint, timeout=timeout)
             sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
         try:
@@ -62,7 +64,24 @@ class _SocketPool:
                 sock.close()
             except OSError:
                 pass
-            raise
+            # a pooled socket may simply be stale (server restarted between
+            # calls): retry ONCE on a fresh connection before giving up — a
+            # fresh connection failing means the server is really down
+            if pooled:
+                sock = socket.create_connection(endpoint, timeout=timeout)
+                sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
+                try:
+                    sock.settimeout(timeout)
+                    send_message(sock, kind, payload)
+                    resp_kind, resp_payload = recv_message(sock)
+                except BaseException:
+                    try:
+                        sock.close()
+                    except OSError:
+                        pass
+                    raise
+            else:
+                raise
         q.put(sock)
         if resp_kind == "error":
             raise RuntimeError(f"remote expert error: {resp_payload.decode(errors='replace')}")

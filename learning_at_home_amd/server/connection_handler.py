@@ -77,6 +77,11 @@ class ConnectionHandler:
                     kind, payload = await aread_message(reader)
                 except (asyncio.IncompleteReadError, ConnectionError):
                     break
+                # requests on one connection are served concurrently, so replies
+                # may come back in any order; the protocol has no request ids,
+                # so clients must keep at most ONE request in flight per
+                # connection (client/_SocketPool checks a socket out exclusively
+                # for the full request/response cycle)
                 asyncio.ensure_future(self._handle_request(kind, payload, writer))
         finally:
             try:

@@ -239,6 +239,42 @@ class TestFaultTolerance:
             dht.shutdown()
 
 
+    def test_stale_pooled_socket_retried(self):
+        """After a server restarts ON THE SAME PORT, a client whose pooled
+        connection went stale must transparently retry on a fresh socket."""
+        import learning_at_home_amd.client.expert as expert_mod
+
+        dht = DHT(start=True)
+        s1 = Server.create(1, "ffn", D_MODEL, expert_prefix="stale",
+                           initial_peers=[dht.endpoint], announce_period=30.0, start=True)
+        port = s1.port
+        try:
+            remote = RemoteExpert("stale.0", "127.0.0.1", port)
+            x = torch.randn(2, D_MODEL)
+            y1 = remote(x)  # populates the socket pool
+            assert y1.shape == (2, D_MODEL)
+            s1.shutdown()  # pooled socket is now stale
+            s2 = Server.create(1, "ffn", D_MODEL, expert_prefix="stale", no_dht=True,
+                               listen_on=("127.0.0.1", port), announce_period=30.0, start=True)
+            try:
+                deadline = time.monotonic() + 10
+                while time.monotonic() < deadline:
+                    try:
+                        y2 = remote(torch.randn(2, D_MODEL))
+                        break
+                    except (OSError, RuntimeError):
+                        time.sleep(0.2)  # port may need a moment to rebind
+                else:
+                    raise AssertionError("retry on fresh socket never succeeded")
+                assert y2.shape == (2, D_MODEL)
+            finally:
+                s2.shutdown()
+        finally:
+            if s1.dht is not None:
+                s1.dht.shutdown()
+            dht.shutdown()
+
+
 class TestBeamSearcher:
     def test_multidim_grid_beam(self, moe_grid):
         """2x2 uid grid 'bs.i.j': beam search finds the live best-scoring
