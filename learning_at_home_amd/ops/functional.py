@@ -240,6 +240,19 @@ class _BatchedFFNHip(torch.autograd.Function):
 
 
 @torch.no_grad()
+def flash_attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
+    """Causal attention forward, [B, H, S, 64] bf16 (EXPERIMENTAL: inference
+    only — no backward yet; training paths keep torch sdpa). On GPU this is
+    the hand-written CDNA4 flash kernel (ops/hip/attention.hip: online
+    softmax, MFMA QK^T and PV, one wave per 16 query rows); elsewhere it
+    falls back to torch sdpa with identical causal semantics."""
+    if q.is_cuda and HAS_HIP and q.dtype == torch.bfloat16 and q.shape[-1] == 64 \
+            and q.shape[-2] % 64 == 0:
+        return _C.flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous())
+    return F.scaled_dot_product_attention(q, k, v, is_causal=True)
+
+
+@torch.no_grad()
 def fused_ffn_infer(x: torch.Tensor, fc1_weight: torch.Tensor, fc1_bias: torch.Tensor,
                     fc2_weight: torch.Tensor, fc2_bias: torch.Tensor) -> torch.Tensor:
     """Inference-only fused FFN for ONE expert (the serving hot path:

@@ -10,6 +10,8 @@ void lah_batched_gemm_nt(const void*, const void*, void*, const void*, void*, in
                          int, int, int, hipStream_t);
 void lah_batched_gemm_tn(const void*, const void*, void*, int, int, int, int, int,
                          hipStream_t);
+void lah_flash_attn_fwd(const void*, const void*, const void*, void*, int, int,
+                        hipStream_t);
 void lah_gating_topk_fwd(const void*, void*, void*, void*, int, int, int, hipStream_t);
 void lah_gating_topk_bwd(const void*, const void*, const void*, const void*, const void*,
                          void*, int, int, int, hipStream_t);
@@ -217,12 +219,28 @@ torch::Tensor colsum(torch::Tensor d) {
 
 }  // namespace
 
+
+// causal flash-attention forward (experimental): q,k,v [B,H,S,64] bf16
+torch::Tensor flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v) {
+  CHECK_CUDA(q); CHECK_BF16(q); CHECK_CONTIG(q);
+  CHECK_BF16(k); CHECK_CONTIG(k); CHECK_BF16(v); CHECK_CONTIG(v);
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == 64, "q must be [B, H, S, 64]");
+  TORCH_CHECK(q.sizes() == k.sizes() && q.sizes() == v.sizes(), "q/k/v shape mismatch");
+  int S = q.size(2);
+  TORCH_CHECK(S % 64 == 0, "seq_len must be a multiple of 64");
+  auto o = torch::empty_like(q);
+  lah_flash_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                     q.size(0) * q.size(1), S, cur_stream());
+  return o;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batched_gemm_nt", &batched_gemm_nt, "batched NT GEMM with fused epilogue",
         py::arg("A"), py::arg("B"), py::arg("bias") = py::none(),
         py::arg("pre") = py::none(), py::arg("epi") = 0);
   m.def("batched_gemm_tn", &batched_gemm_tn, py::arg("A"), py::arg("B"), py::arg("C"),
         py::arg("accumulate") = false);
+  m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("gating_topk_fwd", &gating_topk_fwd);
   m.def("gating_topk_bwd", &gating_topk_bwd, py::arg("probs"), py::arg("topk_idx"),
         py::arg("topk_w"), py::arg("d_topk_w"), py::arg("aux_g") = py::none());

@@ -343,3 +343,21 @@ class TestMoEOnGPU:
         moe_gpu.experts.init_gpu_state()
         y_gpu = moe_gpu(x.to(DEV))
         assert_close_bf16(y_gpu, y_cpu, rtol=5e-2, what="moe gpu vs cpu")
+
+
+def test_flash_attention_fwd_vs_sdpa():
+    """Experimental causal flash-attn forward vs fp32 torch sdpa."""
+    import torch.nn.functional as F
+
+    from learning_at_home_amd.ops.functional import flash_attention_fwd
+
+    torch.manual_seed(0)
+    for B, H, S in ((2, 4, 256), (1, 2, 64), (1, 1, 512)):
+        q = torch.randn(B, H, S, 64, device="cuda:0", dtype=torch.bfloat16)
+        k = torch.randn(B, H, S, 64, device="cuda:0", dtype=torch.bfloat16)
+        v = torch.randn(B, H, S, 64, device="cuda:0", dtype=torch.bfloat16)
+        out = flash_attention_fwd(q, k, v)
+        ref = F.scaled_dot_product_attention(q.float(), k.float(), v.float(), is_causal=True)
+        err = (out.float() - ref).abs().max().item()
+        assert out.shape == q.shape
+        assert err < 3e-2, f"B{B} H{H} S{S}: max err {err}"
