@@ -241,14 +241,45 @@ class _BatchedFFNHip(torch.autograd.Function):
 
 @torch.no_grad()
 def flash_attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
-    """Causal attention forward, [B, H, S, 64] bf16 (EXPERIMENTAL: inference
-    only — no backward yet; training paths keep torch sdpa). On GPU this is
-    the hand-written CDNA4 flash kernel (ops/hip/attention.hip: online
-    softmax, MFMA QK^T and PV, one wave per 16 query rows); elsewhere it
-    falls back to torch sdpa with identical causal semantics."""
+    """Causal attention forward, [B, H, S, 64] bf16 (inference entry point).
+    On GPU this is the hand-written CDNA4 flash kernel (ops/hip/attention.hip:
+    online softmax, MFMA QK^T and PV, one wave per 16 query rows); elsewhere
+    it falls back to torch sdpa with identical causal semantics."""
     if q.is_cuda and HAS_HIP and q.dtype == torch.bfloat16 and q.shape[-1] == 64 \
             and q.shape[-2] % 64 == 0:
         return _C.flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous())
+    return F.scaled_dot_product_attention(q, k, v, is_causal=True)
+
+
+class _FlashAttention(torch.autograd.Function):
+    """Differentiable causal flash attention (EXPERIMENTAL — numerics-verified,
+    not yet wired into the training model; perf tiling is round-2 work).
+    Backward is recompute-based from the saved log-sum-exp: two kernels
+    (dQ over query blocks; dK/dV over key blocks), no atomics; the row term
+    D = rowsum(dO*O) is computed here in fp32 torch (one cheap reduction)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v):
+        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        o, lse = _C.flash_attn_fwd_lse(q, k, v)
+        ctx.save_for_backward(q, k, v, o, lse)
+        return o
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, o, lse = ctx.saved_tensors
+        dout = dout.contiguous()
+        d = (dout.float() * o.float()).sum(dim=-1)  # [B, H, S] fp32
+        dq, dk, dv = _C.flash_attn_bwd(q, k, v, dout, lse, d)
+        return dq, dk, dv
+
+
+def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.Tensor:
+    """Differentiable causal attention. CDNA4 flash kernels when the shapes
+    qualify ([B, H, S, 64] bf16 on GPU, S % 64 == 0), torch sdpa otherwise."""
+    if q.is_cuda and HAS_HIP and q.dtype == torch.bfloat16 and q.shape[-1] == 64 \
+            and q.shape[-2] % 64 == 0:
+        return _FlashAttention.apply(q, k, v)
     return F.scaled_dot_product_attention(q, k, v, is_causal=True)
 
 

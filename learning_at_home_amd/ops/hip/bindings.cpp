@@ -10,8 +10,10 @@ void lah_batched_gemm_nt(const void*, const void*, void*, const void*, void*, in
                          int, int, int, hipStream_t);
 void lah_batched_gemm_tn(const void*, const void*, void*, int, int, int, int, int,
                          hipStream_t);
-void lah_flash_attn_fwd(const void*, const void*, const void*, void*, int, int,
+void lah_flash_attn_fwd(const void*, const void*, const void*, void*, void*, int, int,
                         hipStream_t);
+void lah_flash_attn_bwd(const void*, const void*, const void*, const void*, const void*,
+                        const void*, void*, void*, void*, int, int, hipStream_t);
 void lah_gating_topk_fwd(const void*, void*, void*, void*, int, int, int, hipStream_t);
 void lah_gating_topk_bwd(const void*, const void*, const void*, const void*, const void*,
                          void*, int, int, int, hipStream_t);
@@ -221,17 +223,48 @@ torch::Tensor colsum(torch::Tensor d) {
 
 
 // causal flash-attention forward (experimental): q,k,v [B,H,S,64] bf16
-torch::Tensor flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v) {
+static void check_attn_args(const torch::Tensor& q, const torch::Tensor& k,
+                            const torch::Tensor& v) {
   CHECK_CUDA(q); CHECK_BF16(q); CHECK_CONTIG(q);
   CHECK_BF16(k); CHECK_CONTIG(k); CHECK_BF16(v); CHECK_CONTIG(v);
   TORCH_CHECK(q.dim() == 4 && q.size(3) == 64, "q must be [B, H, S, 64]");
   TORCH_CHECK(q.sizes() == k.sizes() && q.sizes() == v.sizes(), "q/k/v shape mismatch");
-  int S = q.size(2);
-  TORCH_CHECK(S % 64 == 0, "seq_len must be a multiple of 64");
+  TORCH_CHECK(q.size(2) % 64 == 0, "seq_len must be a multiple of 64");
+}
+
+torch::Tensor flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v) {
+  check_attn_args(q, k, v);
   auto o = torch::empty_like(q);
-  lah_flash_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                     q.size(0) * q.size(1), S, cur_stream());
+  lah_flash_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(), nullptr,
+                     q.size(0) * q.size(1), q.size(2), cur_stream());
   return o;
+}
+
+// (o, lse) — the training entry point; lse feeds the recompute-based backward
+std::vector<torch::Tensor> flash_attn_fwd_lse(torch::Tensor q, torch::Tensor k,
+                                              torch::Tensor v) {
+  check_attn_args(q, k, v);
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({q.size(0), q.size(1), q.size(2)},
+                          q.options().dtype(at::kFloat));
+  lah_flash_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
+                     lse.data_ptr(), q.size(0) * q.size(1), q.size(2), cur_stream());
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor q, torch::Tensor k,
+                                          torch::Tensor v, torch::Tensor dout,
+                                          torch::Tensor lse, torch::Tensor d) {
+  check_attn_args(q, k, v);
+  CHECK_BF16(dout); CHECK_CONTIG(dout);
+  CHECK_F32(lse); CHECK_CONTIG(lse); CHECK_F32(d); CHECK_CONTIG(d);
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  lah_flash_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), dout.data_ptr(),
+                     lse.data_ptr(), d.data_ptr(), dq.data_ptr(), dk.data_ptr(),
+                     dv.data_ptr(), q.size(0) * q.size(1), q.size(2), cur_stream());
+  return {dq, dk, dv};
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -241,6 +274,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("batched_gemm_tn", &batched_gemm_tn, py::arg("A"), py::arg("B"), py::arg("C"),
         py::arg("accumulate") = false);
   m.def("flash_attn_fwd", &flash_attn_fwd);
+  m.def("flash_attn_fwd_lse", &flash_attn_fwd_lse);
+  m.def("flash_attn_bwd", &flash_attn_bwd);
   m.def("gating_topk_fwd", &gating_topk_fwd);
   m.def("gating_topk_bwd", &gating_topk_bwd, py::arg("probs"), py::arg("topk_idx"),
         py::arg("topk_w"), py::arg("d_topk_w"), py::arg("aux_g") = py::none());

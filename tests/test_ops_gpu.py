@@ -361,3 +361,24 @@ def test_flash_attention_fwd_vs_sdpa():
         err = (out.float() - ref).abs().max().item()
         assert out.shape == q.shape
         assert err < 3e-2, f"B{B} H{H} S{S}: max err {err}"
+
+
+def test_flash_attention_bwd_vs_sdpa():
+    """Differentiable flash attention: grads vs fp32 torch sdpa autograd."""
+    import torch.nn.functional as F
+
+    from learning_at_home_amd.ops.functional import flash_attention
+
+    torch.manual_seed(1)
+    for B, H, S in ((2, 2, 256), (1, 1, 128)):
+        qkv = [torch.randn(B, H, S, 64, device="cuda:0", dtype=torch.bfloat16,
+                           requires_grad=True) for _ in range(3)]
+        out = flash_attention(*qkv)
+        do = torch.randn_like(out)
+        out.backward(do)
+        ref_qkv = [t.detach().float().requires_grad_(True) for t in qkv]
+        ref = F.scaled_dot_product_attention(*ref_qkv, is_causal=True)
+        ref.backward(do.float())
+        for name, t, r in zip("qkv", qkv, ref_qkv):
+            err = (t.grad.float() - r.grad).abs().max().item()
+            assert err < 6e-2, f"B{B} H{H} S{S} d{name}: max err {err}"
