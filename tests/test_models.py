@@ -34,3 +34,31 @@ class TestChunkedCE:
         loss.backward()
         assert torch.isfinite(loss)
         assert model.embed.weight.grad is not None  # tied lm_head got grads
+
+
+def test_flash_attention_cpu_fallback_matches_sdpa():
+    """On CPU (or non-qualifying shapes) flash_attention must be exactly the
+    torch sdpa causal path, differentiable."""
+    import torch.nn.functional as F
+
+    from learning_at_home_amd.ops.functional import flash_attention, flash_attention_fwd
+
+    torch.manual_seed(0)
+    q = torch.randn(2, 2, 32, 16, requires_grad=True)
+    k = torch.randn(2, 2, 32, 16, requires_grad=True)
+    v = torch.randn(2, 2, 32, 16, requires_grad=True)
+    out = flash_attention(q, k, v)
+    ref = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+    assert torch.equal(out, ref)
+    out.sum().backward()
+    assert q.grad is not None and torch.isfinite(q.grad).all()
+    with torch.no_grad():
+        assert torch.allclose(flash_attention_fwd(q.detach(), k.detach(), v.detach()), ref)
+
+
+def test_can_fuse_ffn_infer_gates_on_cpu():
+    from learning_at_home_amd.models import FFNExpert
+    from learning_at_home_amd.ops.functional import can_fuse_ffn_infer
+
+    e = FFNExpert(128, 256)
+    assert not can_fuse_ffn_infer(torch.randn(4, 128), e)  # CPU / fp32 -> no fuse
