@@ -60,12 +60,23 @@ class DMoETransformerBlock(nn.Module):
         self.ln2 = nn.LayerNorm(d)
         self.moe = moe
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, kv_cache: Optional[list] = None) -> torch.Tensor:
+        """kv_cache (decode only): a 2-element list [k, v] (or empty, filled on
+        prefill) of [B, H, T_past, Dh]; new keys/values are appended and the
+        query attends over the whole cache (is_causal only without a past)."""
         B, S, D = x.shape
         h = self.ln1(x)
         qkv = self.qkv(h).reshape(B, S, 3, self.n_heads, D // self.n_heads)
         q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
-        attn = F.scaled_dot_product_attention(q, k, v, is_causal=True)
+        if kv_cache is not None:
+            if kv_cache:
+                k = torch.cat([kv_cache[0], k], dim=2)
+                v = torch.cat([kv_cache[1], v], dim=2)
+            kv_cache[:] = [k, v]
+            causal = k.shape[2] == S  # prefill; decode steps see only the past
+        else:
+            causal = True
+        attn = F.scaled_dot_product_attention(q, k, v, is_causal=causal)
         x = x + self.proj(attn.transpose(1, 2).reshape(B, S, D))
         x = x + self.moe(self.ln2(x))
         return x
@@ -102,6 +113,36 @@ class TransformerDMoE(nn.Module):
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
         return self.lm_head(self.hidden_states(input_ids))
+
+    @torch.no_grad()
+    def generate(self, input_ids: torch.Tensor, max_new_tokens: int,
+                 temperature: float = 0.0) -> torch.Tensor:
+        """Greedy (temperature=0) or sampled decode with per-block KV caches.
+        Returns input_ids extended by up to max_new_tokens (bounded by the
+        positional table, config.seq_len). Serving entry point — the MoE
+        layers run their normal routed forward per decode step."""
+        B, S = input_ids.shape
+        caches = [[] for _ in self.blocks]
+
+        def step(ids: torch.Tensor, pos0: int) -> torch.Tensor:
+            pos = torch.arange(pos0, pos0 + ids.shape[1], device=ids.device)
+            x = self.embed(ids) + self.pos_embed(pos)[None]
+            for block, cache in zip(self.blocks, caches):
+                x = block(x, kv_cache=cache)
+            logits = self.lm_head(self.ln_f(x[:, -1]))
+            if temperature > 0:
+                probs = F.softmax(logits.float() / temperature, dim=-1)
+                return torch.multinomial(probs, 1)
+            return logits.argmax(dim=-1, keepdim=True)
+
+        out = input_ids
+        nxt = step(input_ids, 0)  # prefill
+        for i in range(max_new_tokens):
+            out = torch.cat([out, nxt], dim=1)
+            if out.shape[1] >= self.config.seq_len or i == max_new_tokens - 1:
+                break
+            nxt = step(nxt, out.shape[1] - 1)
+        return out
 
     def loss(self, input_ids: torch.Tensor, labels: torch.Tensor,
              fused_ce: bool = False) -> torch.Tensor:
