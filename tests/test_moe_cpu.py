@@ -160,6 +160,31 @@ class TestRemoteMixtureOfExperts:
         (y.sum() + moe.aux_loss).backward()
 
 
+class TestDecentralizedInference:
+    def test_generate_through_remote_experts(self, moe_grid):
+        """Decode end-to-end over the decentralized substrate: a tiny
+        Transformer-DMoE whose MoE layers are RemoteMixtureOfExperts backed by
+        the socket/DHT expert grid — every decode step routes through RPC."""
+        from learning_at_home_amd.models import TransformerDMoE, TransformerDMoEConfig
+
+        dht, servers, client_dht = moe_grid
+        torch.manual_seed(0)
+        cfg = TransformerDMoEConfig(vocab_size=64, d_model=D_MODEL, n_heads=4,
+                                    n_layers=1, seq_len=16, num_experts=4, k_best=2,
+                                    ffn_dim=4 * D_MODEL)
+
+        def moe_factory(i, c):
+            return RemoteMixtureOfExperts(
+                in_features=c.d_model, grid_size=(4,), dht=client_dht,
+                uid_prefix="ffn", k_best=2,
+            )
+
+        model = TransformerDMoE(cfg, moe_factory).eval()
+        out = model.generate(torch.randint(0, 64, (1, 4)), max_new_tokens=4)
+        assert out.shape == (1, 8)
+        assert int(out.max()) < 64 and int(out.min()) >= 0
+
+
 class TestFaultTolerance:
     def test_moe_survives_dead_server(self):
         """Kill one of two servers: MoE keeps working on surviving experts
