@@ -191,6 +191,8 @@ def main():
     from learning_at_home_amd.utils.monitor import StageTimers, expert_load_stats
 
     timers = StageTimers() if args.metrics else None
+    if timers is not None:  # per-stage MoE observability on the first layer
+        model.blocks[0].moe.stage_timers = timers
 
     def train_step(step_idx):
         from contextlib import nullcontext

@@ -258,6 +258,9 @@ class ExpertParallelMoE(nn.Module):
         self.experts = ExpertStack(self.num_local, d_model, ffn_dim)
         self.aux_loss: Optional[torch.Tensor] = None
         self.last_counts: Optional[torch.Tensor] = None
+        # observability (SURVEY §5 tracing): attach a utils.monitor.StageTimers
+        # to time gate / dispatch / a2a / expert_ffn / combine separately
+        self.stage_timers = None
         self._capacity_cache: dict = {}
         # elastic membership (parallel/elastic.py): member_slots = ordered
         # alive rank slots; None = static full membership. Dead slots'
@@ -294,6 +297,9 @@ class ExpertParallelMoE(nn.Module):
         return self._capacity_cache[key]
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from contextlib import nullcontext
+
+        tm = self.stage_timers if self.stage_timers is not None else (lambda s: nullcontext())
         lead_shape = x.shape[:-1]
         D = x.shape[-1]
         flat = x.reshape(-1, D)
@@ -305,7 +311,8 @@ class ExpertParallelMoE(nn.Module):
         members = self.member_slots if elastic else list(range(W))
         P = self.num_experts // self.total_slots  # experts per slot (fixed)
 
-        logits = self.gate(flat).float()
+        with tm("moe_gate"):
+            logits = self.gate(flat).float()
         if elastic and not bool(self.expert_alive.all()):
             # dead slots' experts leave the routing (reference: dead experts
             # vanish until their server re-announces)
@@ -315,9 +322,11 @@ class ExpertParallelMoE(nn.Module):
             # keeps receiving tokens early on, preventing rich-get-richer
             # collapse before the balance loss can act
             logits = logits + self.router_noise_std * torch.randn_like(logits)
-        topk_idx, topk_w, probs = Fops.gating_topk(logits, k)
-        with torch.no_grad():
+        with tm("moe_route"):
+            topk_idx, topk_w, probs = Fops.gating_topk(logits, k)
+        with torch.no_grad(), tm("moe_assign"):
             pos, counts, inv = Fops.assign_slots(topk_idx, E, C)
+        with torch.no_grad():
             if elastic:
                 # belt-and-braces: any slot that still picked a dead expert
                 # (e.g. every candidate masked) is treated as dropped
@@ -326,7 +335,8 @@ class ExpertParallelMoE(nn.Module):
         self.last_counts = counts
 
         compute_dtype = torch.bfloat16 if flat.is_cuda else flat.dtype
-        xbuf = Fops.dispatch(flat.to(compute_dtype), topk_idx, pos, inv, k)  # [E, C, D]
+        with tm("moe_dispatch"):
+            xbuf = Fops.dispatch(flat.to(compute_dtype), topk_idx, pos, inv, k)  # [E, C, D]
 
         if W > 1 or elastic:
             if elastic and members != list(range(self.total_slots)):
@@ -334,7 +344,8 @@ class ExpertParallelMoE(nn.Module):
             else:
                 xsend = xbuf
             if W > 1:
-                xsend = _AllToAll.apply(xsend, self.group)  # [W*P, C, D] recv
+                with tm("moe_a2a_fwd"):
+                    xsend = _AllToAll.apply(xsend, self.group)  # [W*P, C, D] recv
             xrecv = (
                 xsend.view(W, P, C, D).permute(1, 0, 2, 3)
                 .reshape(P, W * C, D).contiguous()
@@ -342,7 +353,8 @@ class ExpertParallelMoE(nn.Module):
         else:
             xrecv = xbuf
 
-        ybuf_local = Fops.batched_ffn(xrecv, self.experts)
+        with tm("moe_expert_ffn"):
+            ybuf_local = Fops.batched_ffn(xrecv, self.experts)
 
         if W > 1 or elastic:
             ysend = (
@@ -350,7 +362,8 @@ class ExpertParallelMoE(nn.Module):
                 .reshape(W * P, C, D).contiguous()
             )
             if W > 1:
-                ysend = _AllToAll.apply(ysend, self.group)
+                with tm("moe_a2a_ret"):
+                    ysend = _AllToAll.apply(ysend, self.group)
             if elastic and members != list(range(self.total_slots)):
                 ybuf = torch.zeros(E, C, D, dtype=ysend.dtype, device=ysend.device)
                 for i, s in enumerate(members):
@@ -360,7 +373,8 @@ class ExpertParallelMoE(nn.Module):
         else:
             ybuf = ybuf_local
 
-        y = Fops.combine(ybuf, topk_idx, pos, topk_w, inv)
+        with tm("moe_combine"):
+            y = Fops.combine(ybuf, topk_idx, pos, topk_w, inv)
 
         aux = None
         if self.aux_loss_coef > 0:

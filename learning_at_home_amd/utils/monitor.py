@@ -88,4 +88,12 @@ def amd_smi_snapshot() -> dict:
         out["rocm_smi"] = json.loads(raw) if raw.strip().startswith("{") else raw[-2000:]
     except Exception as e:  # noqa: BLE001
         out["rocm_smi_error"] = repr(e)
+    try:  # xGMI link state/throughput (multi-GPU boxes; absent counters ignored)
+        raw = subprocess.run(
+            ["/opt/rocm/bin/rocm-smi", "--showtopo", "--json"],
+            capture_output=True, text=True, timeout=10,
+        ).stdout
+        out["xgmi_topo"] = json.loads(raw) if raw.strip().startswith("{") else raw[-2000:]
+    except Exception as e:  # noqa: BLE001
+        out["xgmi_error"] = repr(e)
     return out

@@ -180,3 +180,20 @@ def _elastic_moe_worker(rank, world):
 class TestElasticMoE:
     def test_membership_shrink_and_restore(self):
         mp.spawn(_elastic_moe_worker, args=(2,), nprocs=2, join=True)
+
+
+def test_moe_stage_timers_populate():
+    """SURVEY §5 tracing: attaching StageTimers to an ExpertParallelMoE times
+    gate/route/dispatch/expert_ffn/combine individually."""
+    from learning_at_home_amd.parallel import ExpertParallelMoE
+    from learning_at_home_amd.utils.monitor import StageTimers
+
+    torch.manual_seed(0)
+    moe = ExpertParallelMoE(d_model=32, ffn_dim=64, num_experts=4, k=2)
+    timers = StageTimers(use_gpu_events=False)
+    moe.stage_timers = timers
+    moe(torch.randn(3, 8, 32))
+    s = timers.summary()
+    for stage in ("moe_gate", "moe_route", "moe_assign", "moe_dispatch",
+                  "moe_expert_ffn", "moe_combine"):
+        assert s[stage]["count"] == 1, stage
