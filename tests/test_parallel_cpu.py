@@ -197,3 +197,20 @@ def test_moe_stage_timers_populate():
     for stage in ("moe_gate", "moe_route", "moe_assign", "moe_dispatch",
                   "moe_expert_ffn", "moe_combine"):
         assert s[stage]["count"] == 1, stage
+
+
+def test_gating_topk_cpu_fallback_unit():
+    """Direct unit check of the CPU gating fallback: full softmax probs,
+    torch.topk selection, softmax-over-selected weights, and aux gradient
+    flow (the 2048x-scale bug regression guard lives in the GPU suite; this
+    pins the CPU reference the GPU kernels are tested against)."""
+    torch.manual_seed(0)
+    logits = torch.randn(64, 16, requires_grad=True)
+    topk_idx, topk_w, probs = Fops.gating_topk(logits, k=4)
+    assert torch.allclose(probs.sum(-1), torch.ones(64), atol=1e-5)
+    ref_vals, ref_idx = logits.topk(4, dim=-1)
+    assert torch.equal(topk_idx, ref_idx)
+    assert torch.allclose(topk_w, F.softmax(ref_vals, dim=-1), atol=1e-5)
+    assert torch.allclose(topk_w.sum(-1), torch.ones(64), atol=1e-5)
+    (topk_w.sum() + probs.mean()).backward()
+    assert logits.grad is not None and torch.isfinite(logits.grad).all()
