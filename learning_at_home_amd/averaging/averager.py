@@ -95,6 +95,7 @@ class DecentralizedAverager:
         averaging_expiration: float = 3.0,
         request_timeout: float = 5.0,
         listen_on: Endpoint = ("127.0.0.1", 0),
+        compression: str = "none",
         start: bool = False,
     ):
         self.tensors = list(averaged_tensors)
@@ -104,6 +105,11 @@ class DecentralizedAverager:
         self.averaging_expiration = averaging_expiration
         self.request_timeout = request_timeout
         self.listen_on = listen_on
+        # "bf16": ship butterfly parts and averaged replies as bfloat16 over
+        # the wire (half the bytes; owners still accumulate in fp32) — the
+        # reference's tensor-compression knob. All group peers must agree.
+        assert compression in ("none", "bf16"), compression
+        self.compression = compression
         self._tensor_lock = threading.Lock()
         self._rounds: Dict[str, _RoundState] = {}
         self._rounds_lock = threading.Lock()
@@ -238,7 +244,10 @@ class DecentralizedAverager:
                 if self._rounds.get(group_id) is round_state and round_state.result is None:
                     self._rounds.pop(group_id, None)
             raise TimeoutError("averaging round timed out waiting for parts")
-        return "result", PytorchSerializer.dumps(round_state.result)
+        result = round_state.result
+        if self.compression == "bf16":
+            result = result.bfloat16()
+        return "result", PytorchSerializer.dumps(result)
 
     # ---- client side ---------------------------------------------------
     def _rpc(self, endpoint: Endpoint, kind: str, obj, timeout: Optional[float] = None):
@@ -325,14 +334,18 @@ class DecentralizedAverager:
                 self._rounds[group.group_id] = _RoundState(expected_ranks=-1)
             round_state = self._rounds[group.group_id]
         round_state.set_expected(M)
-        round_state.add_part(rank, parts[rank].clone(), my_weight)
+        own_part = parts[rank].clone()
+        if self.compression == "bf16":  # match what remote peers' parts lost
+            own_part = own_part.bfloat16()
+        round_state.add_part(rank, own_part, my_weight)
 
         try:
             results: List[Optional[torch.Tensor]] = [None] * M
             with ThreadPoolExecutor(max_workers=max(1, M - 1)) as pool:
+                wire = (lambda t: t.bfloat16()) if self.compression == "bf16" else (lambda t: t.clone())
                 futures = {
                     j: pool.submit(self._rpc, group.members[j], "reduce",
-                                   [group.group_id, rank, my_weight, parts[j].clone()])
+                                   [group.group_id, rank, my_weight, wire(parts[j])])
                     for j in range(M) if j != rank
                 }
                 for j, fut in futures.items():
@@ -343,7 +356,7 @@ class DecentralizedAverager:
             if not round_state.ready.wait(self.request_timeout + self.averaging_expiration):
                 raise TimeoutError("own part never completed")
             results[rank] = round_state.result
-            averaged = torch.cat([r for r in results])[:numel]
+            averaged = torch.cat([r.float() for r in results])[:numel]
             with self.get_tensors() as tensors:
                 offset = 0
                 for t in tensors:

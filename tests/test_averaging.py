@@ -139,3 +139,35 @@ class TestRcclAverager:
 
     def test_decentralized_sgd_converges_replicas(self):
         mp.spawn(_dsgd_worker, args=(2,), nprocs=2, join=True)
+
+
+def test_averager_bf16_wire_compression():
+    """compression="bf16" halves wire bytes; the group average must match the
+    exact weighted mean within bf16 tolerance."""
+    dht = DHT(start=True)
+    peer_dht = DHT(initial_peers=[dht.endpoint], start=True)
+    t1 = [torch.full((1000,), 1.0)]
+    t2 = [torch.full((1000,), 3.0)]
+    av1 = DecentralizedAverager(t1, dht, prefix="bfc", target_group_size=2,
+                                averaging_expiration=2.0, compression="bf16", start=True)
+    av2 = DecentralizedAverager(t2, peer_dht, prefix="bfc", target_group_size=2,
+                                averaging_expiration=2.0, compression="bf16", start=True)
+    try:
+        import threading
+
+        results = [None, None]
+        th = [threading.Thread(target=lambda i=i, a=a: results.__setitem__(i, a.step(timeout=20.0)))
+              for i, a in enumerate((av1, av2))]
+        th[0].start()
+        time.sleep(0.2)  # stagger: let one peer become the leader first
+        th[1].start()
+        for t in th:
+            t.join(timeout=30)
+        assert results[0] is not None and results[1] is not None, "round did not complete"
+        for tensors in (t1, t2):
+            assert torch.allclose(tensors[0], torch.full((1000,), 2.0), atol=0.02)
+    finally:
+        av1.shutdown()
+        av2.shutdown()
+        peer_dht.shutdown()
+        dht.shutdown()
