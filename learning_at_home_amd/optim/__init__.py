@@ -8,6 +8,7 @@ from __future__ import annotations
 from typing import Optional, Sequence, Union
 
 import torch
+import torch.distributed as dist
 
 from ..averaging import DecentralizedAverager, RcclAverager
 from ..utils import get_logger
@@ -31,6 +32,7 @@ class DecentralizedSGD:
         target_group_size: int = 8,
         opt_cls=torch.optim.SGD,
         averager: Optional[Union[DecentralizedAverager, RcclAverager]] = None,
+        average_gradients: bool = False,
         **opt_kwargs,
     ):
         params = list(params)
@@ -46,11 +48,27 @@ class DecentralizedSGD:
             )
         else:
             self.averager = None
+        # hivemind CollaborativeOptimizer-style synchronous gradient averaging:
+        # every step, grads are all-reduced (AVG) across the group BEFORE the
+        # local optimizer applies them. RCCL tensor-plane only; grads must keep
+        # stable storage, so zero_grad is forced to set_to_none=False.
+        self.average_gradients = average_gradients
+        if average_gradients:
+            assert dist.is_initialized(), "average_gradients needs torch.distributed"
+        self._grad_averager: Optional[RcclAverager] = None
 
     def zero_grad(self, set_to_none: bool = True) -> None:
+        if self.average_gradients:
+            set_to_none = False  # keep grad storage stable for the fused buffer
         self.opt.zero_grad(set_to_none=set_to_none)
 
     def step(self, closure=None):
+        if self.average_gradients:
+            grads = [p.grad for p in self.params if p.grad is not None]
+            if grads:
+                if self._grad_averager is None or len(self._grad_averager.tensors) != len(grads):
+                    self._grad_averager = RcclAverager(grads, dtype=torch.float32)
+                self._grad_averager.step(async_op=False)
         if self.averager is not None and isinstance(self.averager, RcclAverager):
             self.averager.wait()  # ensure any in-flight round has written back
         if isinstance(self.averager, DecentralizedAverager):

@@ -133,12 +133,33 @@ def _dsgd_worker(rank, world):
     dist.destroy_process_group()
 
 
+def _grad_avg_worker(rank, world):
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29533")
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(7)  # SAME init on every rank
+    model = torch.nn.Linear(4, 4)
+    opt = DecentralizedSGD(model.parameters(), lr=0.05, average_gradients=True)
+    torch.manual_seed(100 + rank)  # DIFFERENT data per rank
+    for _ in range(3):
+        loss = model(torch.randn(8, 4)).pow(2).mean()
+        opt.zero_grad(); loss.backward(); opt.step()
+        # synchronous-DP invariant: replicas agree after EVERY step
+        flat = torch.cat([p.detach().reshape(-1) for p in model.parameters()])
+        gathered = [torch.zeros_like(flat) for _ in range(world)]
+        dist.all_gather(gathered, flat)
+        assert torch.allclose(gathered[0], gathered[1], atol=1e-6), "replicas diverged"
+    dist.destroy_process_group()
+
+
 class TestRcclAverager:
     def test_gloo_two_ranks(self):
         mp.spawn(_rccl_worker, args=(2,), nprocs=2, join=True)
 
     def test_decentralized_sgd_converges_replicas(self):
         mp.spawn(_dsgd_worker, args=(2,), nprocs=2, join=True)
+
+    def test_synchronous_gradient_averaging(self):
+        mp.spawn(_grad_avg_worker, args=(2,), nprocs=2, join=True)
 
 
 def test_averager_bf16_wire_compression():
