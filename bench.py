@@ -280,6 +280,11 @@ def main():
         import sys
 
         print("[metrics] stages:", timers.report(), file=sys.stderr)
+        from learning_at_home_amd.utils.monitor import amd_smi_snapshot
+
+        if use_cuda:
+            snap = amd_smi_snapshot()
+            print("[metrics] smi keys:", sorted(snap.keys()), file=sys.stderr)
         counts = model.blocks[0].moe.last_counts
         if counts is not None:
             cap = model.blocks[0].moe.capacity(tokens_per_step_per_rank)
