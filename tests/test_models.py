@@ -116,5 +116,5 @@ def test_cpu_memorization_loss_falls():
     for _ in range(40):
         loss = model.loss(inputs, labels)
         opt.zero_grad(); loss.backward(); opt.step()
-        losses.append(float(loss))
+        losses.append(float(loss.detach()))
     assert losses[-1] < losses[0] - 0.5, f"no learning: {losses[0]:.3f} -> {losses[-1]:.3f}"
