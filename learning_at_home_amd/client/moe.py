@@ -151,7 +151,7 @@ class RemoteMixtureOfExperts(nn.Module):
         endpoint_of = {u: active[u] for row in sel_uids for u in row}
         y = _RemoteMoECallBatched.apply(
             flat, sel_logits, sel_uids, endpoint_of, self.k_min,
-            self.forward_timeout, self.backward_timeout,
+            self.forward_timeout, self.backward_timeout, self.backward_k_min,
         )
 
         if self.aux_loss_coef > 0:
@@ -180,7 +180,8 @@ class _RemoteMoECallBatched(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, sel_logits: torch.Tensor, sel_uids: List[List[str]],
                 endpoint_of: Dict[str, Endpoint], k_min: int,
-                forward_timeout: float, backward_timeout: float):
+                forward_timeout: float, backward_timeout: float,
+                backward_k_min: int = 1):
         N, d = x.shape
         k_max = sel_logits.shape[1]
         x_cpu = x.detach().cpu()
@@ -234,6 +235,7 @@ class _RemoteMoECallBatched(torch.autograd.Function):
         ctx.responded_uids = set(outputs.keys())
         ctx.endpoint_of = endpoint_of
         ctx.backward_timeout = backward_timeout
+        ctx.backward_k_min = backward_k_min
         ctx.device = x.device
         return y.to(x.device)
 
@@ -272,5 +274,15 @@ class _RemoteMoECallBatched(torch.autograd.Function):
                     fut.cancel()
                     logger.warning(f"backward RPC to expert {uid} failed/timed out; zero grads")
 
+        ok = sum(1 for uid, (fut, _) in futures.items()
+                 if fut in done and fut.exception() is None)
+        if ok < min(ctx.backward_k_min, len(live_experts)):
+            # reference semantics: proceed anyway (failed experts contribute
+            # zero grads), but surface that the backward quorum was missed
+            logger.warning(
+                f"backward quorum missed: {ok}/{len(live_experts)} experts "
+                f"returned grads (backward_k_min={ctx.backward_k_min})"
+            )
+
         return (dx.to(ctx.device, dtype=grad_y.dtype),
-                dlogits.to(ctx.device), None, None, None, None, None)
+                dlogits.to(ctx.device), None, None, None, None, None, None)
