@@ -208,7 +208,7 @@ class TestConnectionFraming:
         proto.datagram_received(b"\xff\x00garbage", ("127.0.0.1", 1))  # must not raise
 
 
-class TestMonitor:
+class TestMonitorExtra:
     def test_stage_timers_cpu(self):
         import time as _time
 
