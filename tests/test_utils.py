@@ -235,3 +235,22 @@ class TestMonitorExtra:
         assert skewed["max_load"] == 800 and skewed["min_load"] == 0
         assert skewed["dropped_frac"] == round((800 - 128) / 800, 4)
         assert skewed["entropy_frac"] < 0.01  # all mass on one expert
+
+
+class TestFramingGuards:
+    def test_oversized_frame_rejected(self):
+        """A length prefix beyond MAX_MESSAGE_BYTES must raise, not allocate."""
+        import socket
+
+        from learning_at_home_amd.utils import recv_message
+        from learning_at_home_amd.utils.connection import _LEN, MAX_MESSAGE_BYTES
+
+        a, b = socket.socketpair()
+        try:
+            a.sendall(_LEN.pack(MAX_MESSAGE_BYTES + 1))
+            b.settimeout(5.0)
+            with pytest.raises(ValueError, match="too large"):
+                recv_message(b)
+        finally:
+            a.close()
+            b.close()
