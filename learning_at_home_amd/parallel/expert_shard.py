@@ -18,7 +18,7 @@ fused SGD kernel.
 from __future__ import annotations
 
 import math
-from typing import Optional
+from typing import Optional, Tuple
 
 import torch
 import torch.distributed as dist
@@ -208,6 +208,74 @@ class FusedExpertSGD:
         st.b2_f32.copy_(st.W2_bias.data)
         st.zero_main_grads()
         self.step_count += 1
+
+
+@torch.no_grad()
+def adam_update_(p: torch.Tensor, g: torch.Tensor, m: torch.Tensor, v: torch.Tensor,
+                 step: int, lr: float, beta1: float, beta2: float, eps: float,
+                 weight_decay: float) -> None:
+    """One fp32 Adam update, in place — torch.optim.Adam semantics
+    (decoupled=False: weight decay added to the gradient). Factored out so
+    the math is unit-tested against torch.optim.Adam on CPU."""
+    if weight_decay != 0.0:
+        g = g.add(p, alpha=weight_decay)
+    m.mul_(beta1).add_(g, alpha=1.0 - beta1)
+    v.mul_(beta2).addcmul_(g, g, value=1.0 - beta2)
+    bc1 = 1.0 - beta1 ** step
+    bc2 = 1.0 - beta2 ** step
+    denom = (v / bc2).sqrt_().add_(eps)
+    p.addcdiv_(m, denom, value=-lr / bc1)
+
+
+class ExpertAdam:
+    """Asynchronous per-shard Adam over an ExpertStack (the optimizer-choice
+    counterpart of FusedExpertSGD; SURVEY.md §2 hot-op #5 names "SGD/Adam").
+    Torch-level (hipBLASLt-free elementwise ops) on the fp32 masters, then
+    re-quantizes the bf16 compute copies and their transposes — the default
+    remains FusedExpertSGD, which fuses all of that into one kernel pass.
+    Requires init_gpu_state(master_weights=True) on GPU."""
+
+    def __init__(self, stack: ExpertStack, lr: float = 1e-3,
+                 betas: Tuple[float, float] = (0.9, 0.999), eps: float = 1e-8,
+                 weight_decay: float = 0.0):
+        assert getattr(stack, "master_weights", True), (
+            "ExpertAdam needs fp32 masters: init_gpu_state(master_weights=True)")
+        self.stack = stack
+        self.lr, self.betas, self.eps, self.weight_decay = lr, betas, eps, weight_decay
+        self.step_count = 0
+        self._state = {}  # name -> (m, v)
+
+    def _mv(self, name: str, like: torch.Tensor):
+        if name not in self._state:
+            self._state[name] = (torch.zeros_like(like), torch.zeros_like(like))
+        return self._state[name]
+
+    @torch.no_grad()
+    def step(self) -> None:
+        st = self.stack
+        if not st.grad_ready:
+            return
+        self.step_count += 1
+        b1, b2 = self.betas
+        for name, p, g in (("W1", st.W1.data, st.main_grad_W1),
+                           ("W2", st.W2.data, st.main_grad_W2),
+                           ("b1", st.W1_bias.data, st.main_grad_b1),
+                           ("b2", st.W2_bias.data, st.main_grad_b2)):
+            m, v = self._mv(name, p)
+            adam_update_(p, g, m, v, self.step_count, self.lr, b1, b2,
+                         self.eps, self.weight_decay)
+        # refresh bf16 compute copies + transposes + fp32 bias mirrors
+        st.W1c.copy_(st.W1.data.bfloat16())
+        st.W2c.copy_(st.W2.data.bfloat16())
+        if st.W1c.is_cuda and HAS_HIP:
+            st.W1t.copy_(_C.transpose_bf16(st.W1c))
+            st.W2t.copy_(_C.transpose_bf16(st.W2c))
+        else:
+            st.W1t.copy_(st.W1c.transpose(1, 2).contiguous())
+            st.W2t.copy_(st.W2c.transpose(1, 2).contiguous())
+        st.b1_f32.copy_(st.W1_bias.data)
+        st.b2_f32.copy_(st.W2_bias.data)
+        st.zero_main_grads()
 
 
 class _AllToAll(torch.autograd.Function):

@@ -382,3 +382,31 @@ def test_flash_attention_bwd_vs_sdpa():
         for name, t, r in zip("qkv", qkv, ref_qkv):
             err = (t.grad.float() - r.grad).abs().max().item()
             assert err < 6e-2, f"B{B} H{H} S{S} d{name}: max err {err}"
+
+
+def test_expert_adam_gpu_wiring():
+    """ExpertAdam on a GPU ExpertStack: masters updated by exact Adam math,
+    bf16 copies and transposes refreshed consistently."""
+    from learning_at_home_amd.parallel import ExpertAdam, ExpertStack
+
+    torch.manual_seed(0)
+    st = ExpertStack(2, 128, 256).to("cuda:0")
+    st.init_gpu_state(master_weights=True)
+    opt = ExpertAdam(st, lr=1e-2)
+    st.main_grad_W1.normal_()
+    st.main_grad_W2.normal_()
+    st.main_grad_b1.normal_()
+    st.main_grad_b2.normal_()
+    g1 = st.main_grad_W1.clone()
+    w1_before = st.W1.data.clone()
+    st.grad_ready = True
+    opt.step()
+    assert not torch.allclose(st.W1.data, w1_before)  # moved
+    ref = w1_before.clone()
+    m = torch.zeros_like(ref); v = torch.zeros_like(ref)
+    from learning_at_home_amd.parallel.expert_shard import adam_update_
+    adam_update_(ref, g1, m, v, 1, 1e-2, 0.9, 0.999, 1e-8, 0.0)
+    assert torch.allclose(st.W1.data, ref, atol=1e-7)
+    assert torch.equal(st.W1c, st.W1.data.bfloat16())  # copies refreshed
+    assert torch.equal(st.W1t, st.W1c.transpose(1, 2).contiguous())
+    assert not st.grad_ready  # re-armed for the next backward

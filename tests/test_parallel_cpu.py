@@ -214,3 +214,24 @@ def test_gating_topk_cpu_fallback_unit():
     assert torch.allclose(topk_w.sum(-1), torch.ones(64), atol=1e-5)
     (topk_w.sum() + probs.mean()).backward()
     assert logits.grad is not None and torch.isfinite(logits.grad).all()
+
+
+def test_adam_update_matches_torch_adam():
+    """adam_update_ (ExpertAdam's math) must exactly reproduce
+    torch.optim.Adam over several steps, with and without weight decay."""
+    from learning_at_home_amd.parallel.expert_shard import adam_update_
+
+    torch.manual_seed(0)
+    for wd in (0.0, 0.01):
+        p_ref = torch.randn(64, requires_grad=True)
+        p_mine = p_ref.detach().clone()
+        m = torch.zeros_like(p_mine)
+        v = torch.zeros_like(p_mine)
+        opt = torch.optim.Adam([p_ref], lr=1e-2, betas=(0.9, 0.999), eps=1e-8,
+                               weight_decay=wd)
+        for step in range(1, 6):
+            g = torch.randn(64)
+            p_ref.grad = g.clone()
+            opt.step()
+            adam_update_(p_mine, g.clone(), m, v, step, 1e-2, 0.9, 0.999, 1e-8, wd)
+            assert torch.allclose(p_mine, p_ref.detach(), atol=1e-7), (wd, step)
