@@ -322,3 +322,45 @@ class TestBeamSearcher:
         assert len(found) == 3
         scores_out = [sc for _, _, sc in found]
         assert scores_out == sorted(scores_out, reverse=True)
+
+
+class TestServerCLI:
+    def test_cli_serves_and_shuts_down(self):
+        """The operator journey: `python -m learning_at_home_amd.server`
+        joins a DHT, announces, serves a forward RPC, dies cleanly on SIGTERM."""
+        import re
+        import signal
+        import subprocess
+        import sys
+
+        dht = DHT(start=True)
+        client_dht = DHT(initial_peers=[dht.endpoint], start=True)
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "learning_at_home_amd.server",
+             "--num-experts", "1", "--hidden-dim", str(D_MODEL),
+             "--expert-prefix", "cli", "--announce-period", "1",
+             "--initial-peer", f"{dht.endpoint[0]}:{dht.endpoint[1]}",
+             "--device", "cpu"],
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        )
+        try:
+            deadline = time.monotonic() + 60
+            ep = None
+            while time.monotonic() < deadline and ep is None:
+                (ep,) = client_dht.get_experts(["cli.0"])
+                if ep is None:
+                    time.sleep(0.5)
+                assert proc.poll() is None, f"CLI died: {proc.stdout.read()[-2000:]}"
+            assert ep is not None, "CLI server never announced cli.0"
+            remote = RemoteExpert("cli.0", *ep)
+            y = remote(torch.randn(2, D_MODEL))
+            assert y.shape == (2, D_MODEL) and torch.isfinite(y).all()
+        finally:
+            proc.send_signal(signal.SIGTERM)
+            try:
+                proc.wait(timeout=15)
+            except subprocess.TimeoutExpired:
+                proc.kill()
+            client_dht.shutdown()
+            dht.shutdown()
+        assert proc.returncode == 0, f"CLI exit code {proc.returncode}"
