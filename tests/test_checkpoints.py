@@ -94,3 +94,31 @@ class TestExpertStackInterop:
         assert dst.load_checkpoints(tmp_path, "rt") == 2
         assert torch.equal(src.W1.detach(), dst.W1.detach())
         assert torch.equal(src.W2_bias.detach(), dst.W2_bias.detach())
+
+
+class TestServerRestoreOnStart:
+    def test_server_create_restores_from_checkpoint_dir(self, tmp_path):
+        """Restart journey: a server saves per-expert checkpoints; a FRESH
+        Server.create pointed at the same dir serves the SAME weights."""
+        from learning_at_home_amd import Server
+
+        torch.manual_seed(0)
+        s1 = Server.create(1, "ffn", 32, expert_prefix="ck", no_dht=True,
+                           checkpoint_dir=tmp_path, start=True)
+        try:
+            x = torch.randn(3, 32)
+            with torch.no_grad():
+                y1 = s1.expert_backends["ck.0"].expert(x)
+            store_experts(s1.expert_backends, tmp_path)
+        finally:
+            s1.shutdown()
+
+        torch.manual_seed(99)  # different init: restore must override it
+        s2 = Server.create(1, "ffn", 32, expert_prefix="ck", no_dht=True,
+                           checkpoint_dir=tmp_path, start=True)
+        try:
+            with torch.no_grad():
+                y2 = s2.expert_backends["ck.0"].expert(x)
+            assert torch.allclose(y1, y2, atol=1e-6), "weights not restored"
+        finally:
+            s2.shutdown()
