@@ -3,6 +3,8 @@ ordering; ExpertBackend numerics vs calling the module directly)."""
 import threading
 import time
 
+import pytest
+
 import torch
 
 from learning_at_home_amd.models import FFNExpert
@@ -153,3 +155,34 @@ class TestExpertBackendDtypeBoundary:
         x_ref = x.detach().requires_grad_(True)
         expert32(x_ref).backward(dy)
         assert torch.allclose(gx, x_ref.grad, atol=0.05, rtol=0.05)
+
+
+class TestRuntimeFaultIsolation:
+    def test_bad_batch_poisons_only_its_futures(self):
+        """An expert that raises must fail THAT batch's futures and leave the
+        Runtime serving subsequent requests (server survives poison inputs)."""
+
+        class FlakyExpert(torch.nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.scale = torch.nn.Parameter(torch.ones(1))
+
+            def forward(self, x):
+                if bool((x > 9000).any()):
+                    raise RuntimeError("poison batch")
+                return x * self.scale
+
+        expert = FlakyExpert()
+        b = ExpertBackend("flaky", expert, torch.optim.SGD(expert.parameters(), lr=0.1),
+                          args_schema=(BatchTensorDescriptor.make(4),))
+        rt = Runtime({"flaky": b})
+        rt.run_in_background()
+        try:
+            bad = b.forward_pool.submit_task(torch.full((1, 4), 10000.0))
+            with pytest.raises(Exception, match="poison"):
+                bad.result(timeout=5)
+            good = b.forward_pool.submit_task(torch.ones(2, 4))
+            (out,) = good.result(timeout=5)
+            assert torch.allclose(out, torch.ones(2, 4))
+        finally:
+            rt.shutdown()
