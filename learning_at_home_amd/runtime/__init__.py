@@ -54,13 +54,16 @@ class Runtime:
         logger.debug(f"Runtime started with {len(self.pools)} pools on {self.device}")
         while True:
             with self._cond:
-                while not self._shutdown and not any(len(p) for p in self.pools):
-                    self._cond.wait(timeout=0.5)
+                # a pool is eligible once ready(): min_batch_size reached or
+                # its oldest task aged past the pool's batching timeout
+                while not self._shutdown and not any(p.ready() for p in self.pools):
+                    self._cond.wait(timeout=0.05)
                 if self._shutdown:
                     break
-            pool = min(self.pools, key=lambda p: p.priority)
-            if len(pool) == 0:
+            eligible = [p for p in self.pools if p.ready()]
+            if not eligible:
                 continue
+            pool = min(eligible, key=lambda p: p.priority)
             if self.device is not None and self.device.type != "cpu":
                 processed = self._process_on_device(pool)
             else:

@@ -74,6 +74,22 @@ class TaskPool:
         with self._lock:
             return len(self._queue)
 
+    def ready(self) -> bool:
+        """True when a batch should be formed NOW: queued samples reach
+        min_batch_size, or the oldest task has waited `timeout` seconds
+        (latency bound on the batching wait)."""
+        with self._lock:
+            if not self._queue:
+                return False
+            if self.min_batch_size <= 1:
+                return True
+            total = sum(t.batch_size for t in self._queue)
+            if total >= self.min_batch_size:
+                return True
+            import time as _time
+
+            return (_time.monotonic() - self._queue[0].timestamp) >= self.timeout
+
     def load_batch(self) -> Tuple[List[Task], List[torch.Tensor]]:
         """Pop tasks up to max_batch_size samples; concat their args along dim 0."""
         tasks: List[Task] = []

@@ -186,3 +186,31 @@ class TestRuntimeFaultIsolation:
             assert torch.allclose(out, torch.ones(2, 4))
         finally:
             rt.shutdown()
+
+
+class TestMinBatchAccumulation:
+    def test_min_batch_size_waits_then_timeout_flushes(self):
+        """A pool with min_batch_size accumulates small tasks; a lone task is
+        still flushed once the batching timeout elapses (latency bound)."""
+        expert = FFNExpert(8)
+        b = ExpertBackend("mb", expert, torch.optim.SGD(expert.parameters(), lr=0.1),
+                          args_schema=(BatchTensorDescriptor.make(8),))
+        pool = b.forward_pool
+        pool.min_batch_size, pool.timeout = 4, 0.4
+        rt = Runtime({"mb": b})
+        rt.run_in_background()
+        try:
+            t0 = time.monotonic()
+            lone = pool.submit_task(torch.randn(1, 8))
+            lone.result(timeout=5)
+            waited = time.monotonic() - t0
+            assert waited >= 0.3, f"lone task flushed too early ({waited:.3f}s)"
+
+            # enough samples -> processed promptly, no timeout wait
+            t0 = time.monotonic()
+            futs = [pool.submit_task(torch.randn(2, 8)) for _ in range(2)]
+            for f in futs:
+                f.result(timeout=5)
+            assert time.monotonic() - t0 < 0.3, "full batch should not wait"
+        finally:
+            rt.shutdown()
