@@ -52,5 +52,7 @@ class BatchTensorDescriptor(TensorDescriptor):
         return cls((-1, *trailing_dims), dtype, requires_grad)
 
     def make_batch(self, batch_size: int, **kwargs) -> torch.Tensor:
+        # zeros, not empty: this feeds schema-inference dummy forwards, and
+        # uninitialized garbage can be huge/NaN and break arbitrary experts
         shape = (batch_size, *self.shape[1:])
-        return torch.empty(shape, dtype=self.torch_dtype, **kwargs).requires_grad_(self.requires_grad)
+        return torch.zeros(shape, dtype=self.torch_dtype, **kwargs).requires_grad_(self.requires_grad)
