@@ -254,3 +254,13 @@ class TestFramingGuards:
         finally:
             a.close()
             b.close()
+
+
+def test_tesseract_codename_aliases():
+    """The reference's original class names (Tesseract*) must resolve to the
+    same objects as the modern names — drop-in parity for old import paths."""
+    import learning_at_home_amd as lah
+
+    assert lah.TesseractServer is lah.Server
+    assert lah.TesseractNetwork is lah.DHT
+    assert lah.TesseractRuntime is lah.Runtime
