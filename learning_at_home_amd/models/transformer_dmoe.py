@@ -11,12 +11,21 @@ mapping hidden states [B, S, D] → [B, S, D], so the same trunk runs over
 """
 from __future__ import annotations
 
+import os
 from dataclasses import dataclass, field
 from typing import Callable, Optional
 
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
+
+
+def _use_flash_attention() -> bool:
+    """LAH_FLASH_ATTN=1 routes block attention through the CDNA4 flash
+    kernels (ops.functional.flash_attention; falls back to sdpa whenever the
+    shape/dtype doesn't qualify). Default OFF until the kernel's performance
+    pass lands (docs/ROUND2_NOTES.md §2) — numerics are already verified."""
+    return os.environ.get("LAH_FLASH_ATTN", "0") == "1"
 
 
 @dataclass
@@ -76,7 +85,12 @@ class DMoETransformerBlock(nn.Module):
             causal = k.shape[2] == S  # prefill; decode steps see only the past
         else:
             causal = True
-        attn = F.scaled_dot_product_attention(q, k, v, is_causal=causal)
+        if causal and _use_flash_attention():
+            from ..ops.functional import flash_attention
+
+            attn = flash_attention(q, k, v)
+        else:
+            attn = F.scaled_dot_product_attention(q, k, v, is_causal=causal)
         x = x + self.proj(attn.transpose(1, 2).reshape(B, S, D))
         x = x + self.moe(self.ln2(x))
         return x

@@ -118,3 +118,20 @@ def test_cpu_memorization_loss_falls():
         opt.zero_grad(); loss.backward(); opt.step()
         losses.append(float(loss.detach()))
     assert losses[-1] < losses[0] - 0.5, f"no learning: {losses[0]:.3f} -> {losses[-1]:.3f}"
+
+
+def test_flash_attention_env_flag_cpu(monkeypatch):
+    """LAH_FLASH_ATTN=1 routes block attention through flash_attention (sdpa
+    fallback on CPU — outputs identical to the flag-off path)."""
+    from learning_at_home_amd.models import TransformerDMoE, TransformerDMoEConfig
+
+    torch.manual_seed(0)
+    cfg = TransformerDMoEConfig(vocab_size=64, d_model=32, n_heads=2, n_layers=1,
+                                seq_len=8, num_experts=2, k_best=1, ffn_dim=48)
+    model = TransformerDMoE(cfg).eval()
+    ids = torch.randint(0, 64, (2, 8))
+    with torch.no_grad():
+        base = model(ids)
+        monkeypatch.setenv("LAH_FLASH_ATTN", "1")
+        flagged = model(ids)
+    assert torch.equal(base, flagged)
