@@ -81,6 +81,10 @@ def main():
                         "stochastic-rounding SGD, the fast path)")
     p.add_argument("--metrics", action="store_true",
                    help="print per-stage timers + expert-load stats to stderr")
+    p.add_argument("--fused-ce", action="store_true",
+                   help="chunked fused linear+cross-entropy (never materializes "
+                        "the [T, vocab] logits: -46 GB at the flagship config, "
+                        "~3%% slower -- for memory-constrained configs)")
     p.add_argument("--loss-every", type=int, default=0,
                    help="if >0, rank 0 prints the step loss to stderr every N "
                         "timed steps (long-horizon stability monitoring)")
@@ -200,7 +204,7 @@ def main():
         tm = timers if timers is not None else (lambda s: nullcontext())
         inputs, labels = make_batch()
         with tm("forward"):
-            loss = model.loss(inputs, labels)
+            loss = model.loss(inputs, labels, fused_ce=args.fused_ce)
         trunk_opt.zero_grad(set_to_none=True)
         with tm("backward"):
             loss.backward()
