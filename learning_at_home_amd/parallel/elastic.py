@@ -22,6 +22,7 @@ from __future__ import annotations
 
 import datetime
 import os
+import random
 import socket
 import threading
 import time
@@ -146,7 +147,7 @@ class ElasticGroup:
             if self.slot not in alive:
                 alive = sorted(set(alive + [self.slot]))
             if expected is not None and len(alive) < expected:
-                time.sleep(0.25)
+                time.sleep(0.25 + random.random() * 0.25)
                 continue
             if alive[0] == self.slot:  # coordinator
                 gen_no = self.gen + 1
@@ -159,7 +160,7 @@ class ElasticGroup:
             else:
                 published = self._published_gen()
                 if published is None or published[0] <= self.gen or self.slot not in published[1]:
-                    time.sleep(0.25)
+                    time.sleep(0.25 + random.random() * 0.25)
                     continue
                 target = published
             gen_no, slots, port = target
@@ -178,7 +179,11 @@ class ElasticGroup:
                 logger.warning(f"slot {self.slot}: gen {gen_no} join failed ({e!r}); retrying")
                 if dist.is_initialized():
                     dist.destroy_process_group()
-                time.sleep(0.5)
+                # JITTERED backoff: identical retry periods phase-lock the
+                # whole group into a rendezvous livelock (each member is
+                # inside a doomed init exactly when the next gen appears —
+                # observed under churn in experiments/elastic_train.py)
+                time.sleep(0.5 + random.random() * 1.5)
         raise TimeoutError(f"slot {self.slot}: could not form a group")
 
     @property
