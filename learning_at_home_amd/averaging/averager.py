@@ -19,6 +19,7 @@ The RCCL side-stream variant for intra-node GPU groups is rccl.py.
 from __future__ import annotations
 
 import os
+import random
 import socket
 import threading
 import time
@@ -288,7 +289,10 @@ class DecentralizedAverager:
                 self._leader_state = None
                 if group is not None:
                     return self._run_allreduce(group, weight)
-            time.sleep(0.05)
+            # jittered retry: two peers starting simultaneously otherwise
+            # re-collide on the leader key in lock-step (same failure shape
+            # as the elastic-group rendezvous livelock; see elastic.py)
+            time.sleep(0.05 + random.random() * 0.2)
         return None
 
     def _try_lead(self, weight: float) -> bool:
