@@ -144,10 +144,11 @@ def main():
             dht.declare_experts(uids, "127.0.0.1", 40000 + rank, expiration=60.0)
 
             def announce_loop():
+                import random as _r
                 import time as _t
 
                 while True:
-                    _t.sleep(20.0)
+                    _t.sleep(20.0 + _r.random() * 2.0)  # de-sync rank bursts
                     try:
                         dht.declare_experts(uids, "127.0.0.1", 40000 + rank, expiration=60.0)
                     except Exception:  # noqa: BLE001

@@ -81,7 +81,7 @@ class ElasticGroup:
         self.announce()
 
         def loop():
-            while not self._stop.wait(self.lease / 3):
+            while not self._stop.wait(self.lease / 3 + random.random() * self.lease / 10):
                 try:
                     self.announce()
                 except Exception as e:  # noqa: BLE001

@@ -90,7 +90,11 @@ class Server:
             logger.warning(f"expert announce failed: {e!r}")
 
     def _announce_loop(self) -> None:
-        while not self._stop.wait(self.announce_period):
+        import random as _random
+
+        # jittered re-announce: many servers sharing one period otherwise
+        # burst the DHT simultaneously (lease = 2x period absorbs the drift)
+        while not self._stop.wait(self.announce_period * (1.0 + 0.1 * _random.random())):
             self._announce_once()
 
     def shutdown(self) -> None:
