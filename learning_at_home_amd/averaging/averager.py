@@ -140,6 +140,15 @@ class DecentralizedAverager:
     def shutdown(self) -> None:
         self._stop.set()
         if self._server_sock is not None:
+            # closing the listener does NOT wake a thread blocked in accept()
+            # on Linux — poke it with a self-connection first so the accept
+            # loop observes _stop and exits (verified: the accept thread
+            # outlived shutdown forever without this)
+            try:
+                with socket.create_connection(("127.0.0.1", self.port), timeout=1.0):
+                    pass
+            except OSError:
+                pass
             try:
                 self._server_sock.close()
             except OSError:

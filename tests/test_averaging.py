@@ -192,3 +192,26 @@ def test_averager_bf16_wire_compression():
         av2.shutdown()
         peer_dht.shutdown()
         dht.shutdown()
+
+
+def test_averager_shutdown_reaps_accept_thread():
+    """Regression: closing the listener does not wake a blocked accept() on
+    Linux — shutdown must poke it so the server thread actually exits."""
+    dht = DHT(start=True)
+    try:
+        before = {t.name for t in threading.enumerate()}
+        avs = [DecentralizedAverager([torch.ones(4)], dht, prefix=f"reap{i}",
+                                     averaging_expiration=1.0, start=True)
+               for i in range(3)]
+        for a in avs:
+            a.shutdown()
+        deadline = time.monotonic() + 5
+        while time.monotonic() < deadline:
+            leftover = [t.name for t in threading.enumerate()
+                        if t.name == "averager_server" and t.is_alive()]
+            if not leftover:
+                break
+            time.sleep(0.1)
+        assert not leftover, f"accept threads leaked: {leftover}"
+    finally:
+        dht.shutdown()
