@@ -364,3 +364,42 @@ class TestServerCLI:
             client_dht.shutdown()
             dht.shutdown()
         assert proc.returncode == 0, f"CLI exit code {proc.returncode}"
+
+
+class TestConcurrentClients:
+    def test_parallel_fwd_bwd_hammer(self):
+        """8 client threads hammer one server with mixed forward/backward
+        (socket pool checkout + asyncio handler + TaskPool batching under
+        real contention); every reply correct, server keeps serving."""
+        import threading
+
+        dht = DHT(start=True)
+        srv = Server.create(1, "ffn", 64, expert_prefix="hm", no_dht=True, start=True)
+        try:
+            remote = RemoteExpert("hm.0", "127.0.0.1", srv.port)
+            errs = []
+
+            def hammer(i):
+                try:
+                    torch.manual_seed(i)
+                    for j in range(10):
+                        x = torch.randn(4, 64, requires_grad=(j % 2 == 0))
+                        y = remote(x)
+                        assert y.shape == (4, 64) and torch.isfinite(y).all()
+                        if x.requires_grad:
+                            y.sum().backward()
+                            assert x.grad is not None
+                except Exception as e:  # noqa: BLE001
+                    errs.append((i, repr(e)))
+
+            ths = [threading.Thread(target=hammer, args=(i,)) for i in range(8)]
+            for t in ths:
+                t.start()
+            for t in ths:
+                t.join(60)
+            assert not errs, errs[:3]
+            y = remote(torch.randn(2, 64))
+            assert torch.isfinite(y).all()  # still serving after the burst
+        finally:
+            srv.shutdown()
+            dht.shutdown()
