@@ -131,3 +131,37 @@ class TestExpertDiscovery:
             assert fresh.get_experts(["gone.0"])[0] is None
         finally:
             fresh.shutdown()
+
+
+class TestBlacklistRecovery:
+    def test_restarted_peer_reachable_again(self):
+        """A peer that times out gets blacklisted (lookups skip it); once it
+        sends any datagram again it must be un-blacklisted and serve lookups
+        (protocol.py: blacklist cleared on datagram_received)."""
+        seed = DHT(start=True)
+        a = DHT(initial_peers=[seed.endpoint], start=True)
+        try:
+            # store a key through the mesh, kill the seed -> a's next lookups
+            # time out against it and blacklist it
+            assert a.store("bl_key", "v1", get_dht_time() + 60)
+            seed_ep = seed.endpoint
+            seed.shutdown()
+            t0 = time.monotonic()
+            a.get("bl_gone", None)  # traverse hits the dead seed, blacklists it
+            # a fresh peer on the SAME port as the dead seed: once it pings a,
+            # the blacklist entry must clear and lookups reach it again
+            reborn = DHT(listen_on=seed_ep, initial_peers=[("127.0.0.1", a.port)], start=True)
+            try:
+                assert reborn.store("bl_key2", "v2", get_dht_time() + 60)
+                deadline = time.monotonic() + 20
+                val = None
+                while time.monotonic() < deadline:
+                    val, _ = a.get("bl_key2", None)
+                    if val == "v2":
+                        break
+                    time.sleep(0.5)
+                assert val == "v2", "rejoined peer never served lookups again"
+            finally:
+                reborn.shutdown()
+        finally:
+            a.shutdown()
