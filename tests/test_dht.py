@@ -146,7 +146,6 @@ class TestBlacklistRecovery:
             assert a.store("bl_key", "v1", get_dht_time() + 60)
             seed_ep = seed.endpoint
             seed.shutdown()
-            t0 = time.monotonic()
             a.get("bl_gone", None)  # traverse hits the dead seed, blacklists it
             # a fresh peer on the SAME port as the dead seed: once it pings a,
             # the blacklist entry must clear and lookups reach it again
