@@ -35,7 +35,26 @@ def main():
     p.add_argument("--device", default=None, help="host experts on this device (e.g. cuda:0)")
     p.add_argument("--dtype", default=None, choices=["float32", "bfloat16"],
                    help="host experts in this dtype (bfloat16 + GPU = fused MFMA serving)")
+    p.add_argument("--latency-ms", type=float, default=0.0,
+                   help="software WAN emulation: add this one-way delay to every "
+                        "expert RPC (the reference paper's tc-netem throughput "
+                        "study, reproduced without root)")
     args = p.parse_args()
+
+    if args.latency_ms > 0:
+        import learning_at_home_amd.client.expert as _expert_mod
+        import learning_at_home_amd.client.moe as _moe_mod
+
+        _orig = _expert_mod.rpc_call
+
+        def delayed_rpc(endpoint, kind, obj, timeout=30.0):
+            time.sleep(args.latency_ms / 1e3)  # one-way request delay
+            out = _orig(endpoint, kind, obj, timeout)
+            time.sleep(args.latency_ms / 1e3)  # one-way response delay
+            return out
+
+        _expert_mod.rpc_call = delayed_rpc
+        _moe_mod.rpc_call = delayed_rpc
 
     num_experts = args.servers * args.experts_per_server
     seed_dht = DHT(start=True)
