@@ -99,8 +99,16 @@ class TestDecentralizedAverager:
             src.shutdown(); dst.shutdown(); dht2.shutdown(); dht.shutdown()
 
 
+def _free_port() -> int:
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _rccl_worker(rank, world):
-    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29531")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")  # port set by parent
     dist.init_process_group("gloo", rank=rank, world_size=world)
     params = [torch.full((10,), float(rank)), torch.full((3, 3), float(rank * 10))]
     av = RcclAverager(params, dtype=torch.float32)
@@ -113,7 +121,7 @@ def _rccl_worker(rank, world):
 
 
 def _dsgd_worker(rank, world):
-    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29532")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")  # port set by parent
     dist.init_process_group("gloo", rank=rank, world_size=world)
     torch.manual_seed(rank)
     model = torch.nn.Linear(4, 4)
@@ -134,7 +142,7 @@ def _dsgd_worker(rank, world):
 
 
 def _grad_avg_worker(rank, world):
-    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29533")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")  # port set by parent
     dist.init_process_group("gloo", rank=rank, world_size=world)
     torch.manual_seed(7)  # SAME init on every rank
     model = torch.nn.Linear(4, 4)
@@ -153,12 +161,15 @@ def _grad_avg_worker(rank, world):
 
 class TestRcclAverager:
     def test_gloo_two_ranks(self):
+        os.environ["MASTER_PORT"] = str(_free_port())  # avoid fixed-port collisions
         mp.spawn(_rccl_worker, args=(2,), nprocs=2, join=True)
 
     def test_decentralized_sgd_converges_replicas(self):
+        os.environ["MASTER_PORT"] = str(_free_port())
         mp.spawn(_dsgd_worker, args=(2,), nprocs=2, join=True)
 
     def test_synchronous_gradient_averaging(self):
+        os.environ["MASTER_PORT"] = str(_free_port())
         mp.spawn(_grad_avg_worker, args=(2,), nprocs=2, join=True)
 
 

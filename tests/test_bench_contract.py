@@ -4,6 +4,14 @@ import subprocess
 import sys
 
 
+def _free_port() -> int:
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def test_bench_json_contract():
     out = subprocess.run(
         [sys.executable, "bench.py", "--steps", "2", "--warmup", "1"],
@@ -26,7 +34,7 @@ def test_bench_distributed_2proc_gloo():
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29561", "bench.py", "--steps", "2", "--warmup", "1"],
+         "--master-port", str(_free_port()), "bench.py", "--steps", "2", "--warmup", "1"],
         capture_output=True, text=True, timeout=600,
     )
     lines = [ln for ln in out.stdout.strip().splitlines() if ln.startswith("{")]

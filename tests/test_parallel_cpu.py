@@ -12,6 +12,14 @@ from learning_at_home_amd.ops import functional as Fops
 from learning_at_home_amd.parallel import ExpertParallelMoE
 
 
+def _free_port() -> int:
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _manual_moe_reference(x_flat, moe, capacity=None):
     """Dense reference: per-token top-k, softmax over selected, expert FFNs."""
     logits = moe.gate(x_flat).float()
@@ -99,7 +107,7 @@ class TestMoESingleProcess:
 
 
 def _ep_worker(rank, world, tmpdir):
-    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29521")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")  # port set by parent
     dist.init_process_group("gloo", rank=rank, world_size=world)
     # single-rank subgroups so `full` runs all experts locally on each rank
     solo_groups = [dist.new_group([r]) for r in range(world)]
@@ -137,11 +145,12 @@ class TestExpertParallelGloo:
     def test_two_rank_matches_local(self, tmp_path):
         """2-rank expert-parallel forward == single-process all-local forward
         (token dispatch over all_to_all, experts sharded)."""
+        os.environ["MASTER_PORT"] = str(_free_port())
         mp.spawn(_ep_worker, args=(2, str(tmp_path)), nprocs=2, join=True)
 
 
 def _elastic_moe_worker(rank, world):
-    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT="29541")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")  # port set by parent
     dist.init_process_group("gloo", rank=rank, world_size=world)
     solo_groups = [dist.new_group([r]) for r in range(world)]
     torch.manual_seed(7)
@@ -179,6 +188,7 @@ def _elastic_moe_worker(rank, world):
 
 class TestElasticMoE:
     def test_membership_shrink_and_restore(self):
+        os.environ["MASTER_PORT"] = str(_free_port())
         mp.spawn(_elastic_moe_worker, args=(2,), nprocs=2, join=True)
 
 
