@@ -452,3 +452,20 @@ class TestFailureRateTraining:
                 srv.dht.shutdown()
             client_dht.shutdown()
             dht.shutdown()
+
+
+def test_large_tensor_rpc_roundtrip():
+    """A multi-megabyte batch travels the socket framing intact (chunked
+    send/recv, not a single syscall) and grads flow back at the same size."""
+    dht = DHT(start=True)
+    srv = Server.create(1, "ffn", 256, expert_prefix="big", no_dht=True, start=True)
+    try:
+        remote = RemoteExpert("big.0", "127.0.0.1", srv.port)
+        x = torch.randn(10000, 256, requires_grad=True)  # ~10 MB fp32
+        y = remote(x)
+        assert y.shape == (10000, 256) and torch.isfinite(y).all()
+        y.sum().backward()
+        assert x.grad is not None and x.grad.shape == x.shape
+    finally:
+        srv.shutdown()
+        dht.shutdown()
