@@ -8,7 +8,10 @@ and feeds dX / dW GEMMs directly. Library GEMMs (hipBLASLt) — this is a
 torch-level fusion, not a kernel.
 
 Numerics: per-chunk logits in fp32 via bf16 GEMM (same as the unfused path);
-loss is the exact token-mean CE.
+loss is the exact valid-token-mean CE. Labels equal to `ignore_index`
+(default -100, matching F.cross_entropy) contribute neither to the loss
+nor to the gradients, and the mean is over VALID tokens only — identical
+semantics to the unfused `F.cross_entropy(..., ignore_index=-100)` path.
 """
 from __future__ import annotations
 
@@ -18,41 +21,55 @@ import torch
 class _ChunkedLinearCE(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: torch.Tensor, weight: torch.Tensor, labels: torch.Tensor,
-                chunk: int):
+                chunk: int, ignore_index: int):
         T = x.shape[0]
         total = torch.zeros((), dtype=torch.float32, device=x.device)
+        n_valid = int((labels != ignore_index).sum().item())
         for s in range(0, T, chunk):
             xs = x[s : s + chunk]
             logits = (xs @ weight.t()).float()
             total += torch.nn.functional.cross_entropy(
-                logits, labels[s : s + chunk], reduction="sum"
+                logits, labels[s : s + chunk], reduction="sum",
+                ignore_index=ignore_index,
             )
         ctx.save_for_backward(x, weight, labels)
         ctx.chunk = chunk
-        return total / T
+        ctx.ignore_index = ignore_index
+        ctx.n_valid = n_valid
+        return total / max(n_valid, 1)
 
     @staticmethod
     def backward(ctx, dloss: torch.Tensor):
         x, weight, labels = ctx.saved_tensors
         T = x.shape[0]
         chunk = ctx.chunk
+        ignore_index = ctx.ignore_index
         dx = torch.empty_like(x)
         dw = torch.zeros_like(weight, dtype=torch.float32)
-        scale = (dloss.float() / T)
+        scale = dloss.float() / max(ctx.n_valid, 1)
         for s in range(0, T, chunk):
             xs = x[s : s + chunk]
+            ls = labels[s : s + chunk]
+            valid = ls != ignore_index
             logits = (xs @ weight.t()).float()
             probs = torch.softmax(logits, dim=-1)
-            probs[torch.arange(probs.shape[0], device=x.device), labels[s : s + chunk]] -= 1.0
+            # clamp ignored labels to 0 for the scatter (their rows are zeroed
+            # below) — an unclamped -100 would wrap to column V-100
+            safe_labels = torch.where(valid, ls, torch.zeros_like(ls))
+            probs[torch.arange(probs.shape[0], device=x.device), safe_labels] -= 1.0
+            probs[~valid] = 0.0
             dlogits = (probs * scale).to(x.dtype)
             dx[s : s + chunk] = dlogits @ weight
             dw += (dlogits.t() @ xs).float()
-        return dx, dw.to(weight.dtype), None, None
+        return dx, dw.to(weight.dtype), None, None, None
 
 
 def chunked_linear_cross_entropy(x: torch.Tensor, weight: torch.Tensor,
-                                 labels: torch.Tensor, chunk: int = 8192) -> torch.Tensor:
-    """loss = CE(x @ weight^T, labels), token-mean, without materializing the
-    full logits. x [T, D], weight [V, D] (tied lm_head), labels [T]."""
+                                 labels: torch.Tensor, chunk: int = 8192,
+                                 ignore_index: int = -100) -> torch.Tensor:
+    """loss = CE(x @ weight^T, labels), valid-token-mean, without
+    materializing the full logits. x [T, D], weight [V, D] (tied lm_head),
+    labels [T]; labels == ignore_index are skipped exactly as in
+    F.cross_entropy."""
     return _ChunkedLinearCE.apply(x.reshape(-1, x.shape[-1]), weight,
-                                  labels.reshape(-1), chunk)
+                                  labels.reshape(-1), chunk, ignore_index)

@@ -119,13 +119,21 @@ class TaskPool:
             outputs = self.process_func(*batch)
         except Exception as e:  # noqa: BLE001
             for t in tasks:
-                if not t.future.done():
-                    t.future.set_exception(e)
+                try:
+                    if not t.future.done():
+                        t.future.set_exception(e)
+                except RuntimeError:  # cancel() raced the done() check
+                    pass
             return 0
         offset = 0
         for t in tasks:
             part = tuple(out[offset : offset + t.batch_size] for out in outputs)
             offset += t.batch_size
-            if not t.future.done():
-                t.future.set_result(part)
+            # a cancel() racing between done() and set_result() must not kill
+            # the Runtime thread: the client already gave up on this task
+            try:
+                if not t.future.done():
+                    t.future.set_result(part)
+            except RuntimeError:
+                logger.debug(f"task in pool {self.name} finished after client cancel; dropping result")
         return offset

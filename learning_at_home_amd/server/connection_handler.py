@@ -77,12 +77,12 @@ class ConnectionHandler:
                     kind, payload = await aread_message(reader)
                 except (asyncio.IncompleteReadError, ConnectionError):
                     break
-                # requests on one connection are served concurrently, so replies
-                # may come back in any order; the protocol has no request ids,
-                # so clients must keep at most ONE request in flight per
-                # connection (client/_SocketPool checks a socket out exclusively
-                # for the full request/response cycle)
-                asyncio.ensure_future(self._handle_request(kind, payload, writer))
+                # one request at a time per connection: the frame has no
+                # request ids, so replies must come back in request order even
+                # for a client that pipelines. Concurrency comes from separate
+                # connections (client/_SocketPool opens one per in-flight call),
+                # each served by its own asyncio task.
+                await self._handle_request(kind, payload, writer)
         finally:
             try:
                 writer.close()

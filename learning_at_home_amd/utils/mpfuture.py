@@ -53,7 +53,7 @@ class MPFuture:
                 # the sender process exits before the receiver detaches the fd
                 from .serializer import PytorchSerializer
 
-                self._send_conn.send_bytes(PytorchSerializer.dumps(["result", value]))
+                self._send_conn.send_bytes(PytorchSerializer.dumps(["result", value], allow_pickle=True))
             except (BrokenPipeError, OSError):
                 pass  # awaiter is gone (e.g. client timed out): drop silently
         self._result = value
@@ -66,7 +66,7 @@ class MPFuture:
             try:
                 from .serializer import PytorchSerializer
 
-                self._send_conn.send_bytes(PytorchSerializer.dumps(["exception", exc]))
+                self._send_conn.send_bytes(PytorchSerializer.dumps(["exception", exc], allow_pickle=True))
             except (BrokenPipeError, OSError):
                 pass
         self._exception = exc
@@ -85,7 +85,7 @@ class MPFuture:
             if self._recv_conn.poll(remaining):
                 from .serializer import PytorchSerializer
 
-                kind, payload = PytorchSerializer.loads(self._recv_conn.recv_bytes())
+                kind, payload = PytorchSerializer.loads(self._recv_conn.recv_bytes(), allow_pickle=True)
                 if kind == "result":
                     self._result = payload
                 else:

@@ -12,6 +12,7 @@ unavoidable host staging for cross-host RPC).
 """
 from __future__ import annotations
 
+import builtins
 import io
 from typing import Any
 
@@ -20,6 +21,17 @@ import numpy as np
 import torch
 
 _TENSOR_EXT = 0x54  # 'T'
+_EXC_EXT = 0x45  # 'E'
+
+
+class RemoteException(Exception):
+    """An exception raised on a remote peer whose type could not be safely
+    reconstructed locally. Carries the original type name and message."""
+
+    def __init__(self, type_name: str, message: str):
+        super().__init__(f"{type_name}: {message}")
+        self.type_name = type_name
+        self.message = message
 
 # torch dtypes without a numpy equivalent are round-tripped through a raw
 # uint8 view (bfloat16 being the one that matters).
@@ -65,11 +77,30 @@ def _unpack_tensor(data: bytes) -> torch.Tensor:
     return out
 
 
+def _pack_exception(exc: BaseException) -> bytes:
+    return msgpack.packb([type(exc).__name__, str(exc)], use_bin_type=True)
+
+
+def _unpack_exception(data: bytes) -> BaseException:
+    type_name, message = msgpack.unpackb(data, raw=False)
+    # Reconstruct only builtin exception types (no arbitrary class lookup):
+    # a constructor like ValueError(message) runs no user code.
+    cand = getattr(builtins, type_name, None)
+    if isinstance(cand, type) and issubclass(cand, BaseException):
+        try:
+            return cand(message)
+        except Exception:  # exotic builtin signature (e.g. UnicodeDecodeError)
+            pass
+    return RemoteException(type_name, message)
+
+
 def _default(obj: Any):
     if isinstance(obj, torch.Tensor):
         return msgpack.ExtType(_TENSOR_EXT, _pack_tensor(obj))
     if isinstance(obj, np.ndarray):
         return msgpack.ExtType(_TENSOR_EXT, _pack_tensor(torch.from_numpy(obj)))
+    if isinstance(obj, BaseException):
+        return msgpack.ExtType(_EXC_EXT, _pack_exception(obj))
     if isinstance(obj, (np.integer,)):
         return int(obj)
     if isinstance(obj, (np.floating,)):
@@ -80,6 +111,8 @@ def _default(obj: Any):
 def _ext_hook(code: int, data: bytes):
     if code == _TENSOR_EXT:
         return _unpack_tensor(data)
+    if code == _EXC_EXT:
+        return _unpack_exception(data)
     return msgpack.ExtType(code, data)
 
 
@@ -98,25 +131,41 @@ class MSGPackSerializer:
 class PytorchSerializer:
     """Tensors and nested python structures of tensors ⇄ bytes.
 
-    Uses msgpack + raw tensor payloads (not pickle): faster, no arbitrary
-    code execution on loads, and dtype/shape/requires_grad preserved.
-    Falls back to torch.save for objects msgpack can't express.
+    Uses msgpack + raw tensor payloads: no arbitrary code execution on
+    loads, dtype/shape/requires_grad preserved, exceptions round-tripped
+    as (type_name, message) pairs — builtin exception types are
+    reconstructed, anything else becomes RemoteException.
+
+    The pickle (torch.save) path exists ONLY for trusted same-host
+    transports (multiprocessing pipes between our own processes) and is
+    disabled by default: `loads` refuses a pickle payload and `dumps`
+    refuses to produce one unless `allow_pickle=True` is passed
+    explicitly. Network paths (expert RPC, averager sockets, DHT) must
+    never pass it — a pickle payload from a peer is arbitrary code
+    execution.
     """
 
     @staticmethod
-    def dumps(obj: Any) -> bytes:
+    def dumps(obj: Any, *, allow_pickle: bool = False) -> bytes:
         try:
             return b"M" + MSGPackSerializer.dumps(obj)
         except TypeError:
+            if not allow_pickle:
+                raise
             buf = io.BytesIO()
             torch.save(obj, buf)
             return b"P" + buf.getvalue()
 
     @staticmethod
-    def loads(buf: bytes) -> Any:
+    def loads(buf: bytes, *, allow_pickle: bool = False) -> Any:
         tag, body = buf[:1], buf[1:]
         if tag == b"M":
             return MSGPackSerializer.loads(body)
         elif tag == b"P":
+            if not allow_pickle:
+                raise ValueError(
+                    "refusing to unpickle a b'P' payload on an untrusted path; "
+                    "pass allow_pickle=True only for trusted same-host pipes"
+                )
             return torch.load(io.BytesIO(body), weights_only=False)
         raise ValueError(f"unknown serialization tag {tag!r}")

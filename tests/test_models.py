@@ -24,6 +24,37 @@ class TestChunkedCE:
         assert torch.allclose(x.grad, x2.grad, atol=1e-5)
         assert torch.allclose(w.grad, w2.grad, atol=1e-4)
 
+    def test_ignore_index_matches_unfused(self):
+        # padded labels (-100) must contribute neither loss nor gradient,
+        # and the mean must be over valid tokens only
+        torch.manual_seed(2)
+        T, D, V = 41, 16, 50
+        x = torch.randn(T, D, requires_grad=True)
+        w = torch.randn(V, D, requires_grad=True)
+        labels = torch.randint(0, V, (T,))
+        labels[::3] = -100  # ~1/3 padded, spread across chunks
+        loss = chunked_linear_cross_entropy(x, w, labels, chunk=8)
+        loss.backward()
+
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        ref = F.cross_entropy((x2 @ w2.t()).float(), labels, ignore_index=-100)
+        ref.backward()
+        assert torch.allclose(loss, ref, atol=1e-5)
+        assert torch.allclose(x.grad, x2.grad, atol=1e-5)
+        assert torch.allclose(w.grad, w2.grad, atol=1e-4)
+        # ignored tokens get exactly zero dx
+        assert torch.all(x.grad[::3] == 0)
+
+    def test_all_labels_ignored_is_finite(self):
+        x = torch.randn(8, 16, requires_grad=True)
+        w = torch.randn(50, 16, requires_grad=True)
+        labels = torch.full((8,), -100)
+        loss = chunked_linear_cross_entropy(x, w, labels, chunk=4)
+        loss.backward()
+        assert loss.item() == 0.0
+        assert torch.all(x.grad == 0) and torch.all(w.grad == 0)
+
     def test_model_loss_finite_and_trains(self):
         torch.manual_seed(1)
         cfg = TransformerDMoEConfig(vocab_size=128, d_model=32, n_heads=2, n_layers=1,

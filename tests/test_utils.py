@@ -193,12 +193,38 @@ class TestConnectionFraming:
         client.close()
         server.close()
 
-    def test_serializer_pickle_fallback(self):
-        # complex(...) is not msgpack-encodable -> exercises the torch.save path
+    def test_serializer_pickle_fallback_gated(self):
+        # complex(...) is not msgpack-encodable -> the torch.save path requires
+        # explicit allow_pickle on BOTH ends (trusted same-host pipes only)
+        import pytest
+
         from learning_at_home_amd.utils import PytorchSerializer
 
         obj = {"z": complex(1, 2)}
-        assert PytorchSerializer.loads(PytorchSerializer.dumps(obj)) == obj
+        with pytest.raises(TypeError):
+            PytorchSerializer.dumps(obj)  # network-path default: no pickle out
+        blob = PytorchSerializer.dumps(obj, allow_pickle=True)
+        with pytest.raises(ValueError):
+            PytorchSerializer.loads(blob)  # network-path default: no pickle in
+        assert PytorchSerializer.loads(blob, allow_pickle=True) == obj
+
+    def test_serializer_exception_roundtrip_no_pickle(self):
+        # exceptions ride msgpack (type name + message), never pickle
+        from learning_at_home_amd.utils import PytorchSerializer
+        from learning_at_home_amd.utils.serializer import RemoteException
+
+        blob = PytorchSerializer.dumps(["exception", ValueError("boom")])
+        assert blob[:1] == b"M"
+        kind, exc = PytorchSerializer.loads(blob)
+        assert isinstance(exc, ValueError) and str(exc) == "boom"
+
+        class CustomError(Exception):
+            pass
+
+        blob = PytorchSerializer.dumps(CustomError("custom boom"))
+        exc = PytorchSerializer.loads(blob)
+        assert isinstance(exc, RemoteException)
+        assert exc.type_name == "CustomError" and "custom boom" in str(exc)
 
     def test_malformed_dht_datagram_ignored(self):
         from learning_at_home_amd.dht.protocol import DHTProtocol
