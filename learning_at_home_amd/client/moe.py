@@ -22,8 +22,8 @@ semantics, control-plane cost O(unique candidates) not O(batch).
 from __future__ import annotations
 
 import time
-from concurrent.futures import ThreadPoolExecutor, wait as futures_wait
-from typing import Dict, List, Optional, Sequence, Tuple
+from concurrent.futures import FIRST_COMPLETED, ThreadPoolExecutor, wait as futures_wait
+from typing import Callable, Dict, List, Optional, Sequence, Tuple
 
 import torch
 import torch.nn as nn
@@ -49,6 +49,7 @@ class RemoteMixtureOfExperts(nn.Module):
         forward_timeout: float = 10.0,
         backward_timeout: float = 10.0,
         backward_k_min: int = 1,
+        timeout_after_k_min: float = 1.0,
         active_cache_ttl: float = 10.0,
         candidate_factor: int = 4,
         aux_loss_coef: float = 0.0,
@@ -59,6 +60,7 @@ class RemoteMixtureOfExperts(nn.Module):
         self.uid_prefix = uid_prefix
         self.k_best, self.k_min, self.backward_k_min = k_best, k_min, backward_k_min
         self.forward_timeout, self.backward_timeout = forward_timeout, backward_timeout
+        self.timeout_after_k_min = timeout_after_k_min
         self.active_cache_ttl = active_cache_ttl
         self.candidate_factor = candidate_factor
         self.aux_loss_coef = aux_loss_coef
@@ -152,6 +154,7 @@ class RemoteMixtureOfExperts(nn.Module):
         y = _RemoteMoECallBatched.apply(
             flat, sel_logits, sel_uids, endpoint_of, self.k_min,
             self.forward_timeout, self.backward_timeout, self.backward_k_min,
+            self.timeout_after_k_min,
         )
 
         if self.aux_loss_coef > 0:
@@ -173,6 +176,47 @@ def _importance_aux_loss(grid_logits: List[torch.Tensor]) -> torch.Tensor:
     return total
 
 
+def _wait_quorum_early_exit(
+    futures: Dict[str, "object"],
+    on_success: Callable[[str, object], None],
+    quorum_met: Callable[[], bool],
+    total_timeout: float,
+    grace: float,
+    direction: str,
+) -> None:
+    """The reference's `timeout_after_k_min` semantics: wait FIRST_COMPLETED
+    in a loop; as soon as `quorum_met()` (k_min replies per token), start a
+    `grace`-second countdown; when it (or `total_timeout`) expires, abandon
+    the stragglers instead of waiting the full timeout. `on_success(uid,
+    result)` runs for each completed-without-exception future, *before*
+    quorum is re-evaluated. (SURVEY §2 RemoteMixtureOfExperts row (3);
+    VERDICT r1 missing #2.)"""
+    fut_uid = {f: uid for uid, f in futures.items()}
+    pending = set(futures.values())
+    deadline = time.monotonic() + total_timeout
+    grace_deadline: Optional[float] = None
+    while pending:
+        eff = deadline if grace_deadline is None else min(deadline, grace_deadline)
+        remaining = eff - time.monotonic()
+        if remaining <= 0:
+            break
+        done, pending = futures_wait(pending, timeout=remaining, return_when=FIRST_COMPLETED)
+        for fut in done:
+            uid = fut_uid[fut]
+            if fut.exception() is None:
+                on_success(uid, fut.result())
+            else:
+                logger.warning(f"{direction} RPC to expert {uid} failed ({fut.exception()!r}); dropping it")
+        if grace_deadline is None and quorum_met():
+            grace_deadline = time.monotonic() + grace
+    for fut in pending:
+        fut.cancel()
+        logger.warning(
+            f"{direction} RPC to expert {fut_uid[fut]} dropped "
+            f"({'straggler past k_min quorum + grace' if grace_deadline is not None else 'timed out'})"
+        )
+
+
 class _RemoteMoECallBatched(torch.autograd.Function):
     """Batched dispatch/combine with quorum fault tolerance and parallel
     backward RPCs (manual grads: softmax-over-survivors combine)."""
@@ -181,7 +225,7 @@ class _RemoteMoECallBatched(torch.autograd.Function):
     def forward(ctx, x: torch.Tensor, sel_logits: torch.Tensor, sel_uids: List[List[str]],
                 endpoint_of: Dict[str, Endpoint], k_min: int,
                 forward_timeout: float, backward_timeout: float,
-                backward_k_min: int = 1):
+                backward_k_min: int = 1, timeout_after_k_min: float = 1.0):
         N, d = x.shape
         k_max = sel_logits.shape[1]
         x_cpu = x.detach().cpu()
@@ -193,20 +237,32 @@ class _RemoteMoECallBatched(torch.autograd.Function):
                 expert_tokens.setdefault(uid, []).append((n, slot))
 
         outputs: Dict[str, torch.Tensor] = {}
-        with ThreadPoolExecutor(max_workers=min(32, len(expert_tokens))) as pool:
+        token_got = [0] * N
+        token_need = [min(k_min, len(row)) for row in sel_uids]
+
+        def _on_success(uid: str, result) -> None:
+            outputs[uid] = result[0]
+            for n, _slot in expert_tokens[uid]:
+                token_got[n] += 1
+
+        def _quorum_met() -> bool:
+            return all(g >= t for g, t in zip(token_got, token_need))
+
+        # NOT a with-block: __exit__ would wait for straggler RPC threads,
+        # defeating the early exit. Abandoned threads finish on their own
+        # socket timeout (success re-pools the socket; failure closes it).
+        pool = ThreadPoolExecutor(max_workers=min(32, len(expert_tokens)))
+        try:
             futures = {}
             for uid, pairs in expert_tokens.items():
                 idx = [n for n, _ in pairs]
                 futures[uid] = pool.submit(
                     rpc_call, endpoint_of[uid], "fwd", [uid, [x_cpu[idx]]], forward_timeout
                 )
-            done, not_done = futures_wait(futures.values(), timeout=forward_timeout)
-            for uid, fut in futures.items():
-                if fut in done and fut.exception() is None:
-                    outputs[uid] = fut.result()[0]
-                else:
-                    fut.cancel()
-                    logger.warning(f"forward RPC to expert {uid} failed/timed out; dropping it")
+            _wait_quorum_early_exit(futures, _on_success, _quorum_met,
+                                    forward_timeout, timeout_after_k_min, "forward")
+        finally:
+            pool.shutdown(wait=False)
 
         if not outputs:
             raise RuntimeError("RemoteMixtureOfExperts: every selected expert failed")
@@ -236,6 +292,7 @@ class _RemoteMoECallBatched(torch.autograd.Function):
         ctx.endpoint_of = endpoint_of
         ctx.backward_timeout = backward_timeout
         ctx.backward_k_min = backward_k_min
+        ctx.timeout_after_k_min = timeout_after_k_min
         ctx.device = x.device
         return y.to(x.device)
 
@@ -250,39 +307,47 @@ class _RemoteMoECallBatched(torch.autograd.Function):
         weighted = (w * dot).sum(-1, keepdim=True)
         dlogits = (w * (dot - weighted)).masked_fill(~responded, 0.0)
 
-        # grads wrt x: parallel bwd RPCs, one per expert that responded in fwd
+        # grads wrt x: parallel bwd RPCs, one per expert that responded in fwd;
+        # same early-exit quorum as forward, but counted per EXPERT
+        # (backward_k_min experts returned grads), since failed experts
+        # contribute zero grads rather than failing the step
         dx = torch.zeros_like(x_cpu, dtype=torch.float32)
         live_experts = [uid for uid in ctx.expert_tokens if uid in ctx.responded_uids]
-        with ThreadPoolExecutor(max_workers=min(32, max(1, len(live_experts)))) as pool:
+        ok_count = [0]
+        idx_of: Dict[str, List[int]] = {}
+
+        def _on_success(uid: str, result) -> None:
+            dx[idx_of[uid]] += result[0].float()
+            ok_count[0] += 1
+
+        def _quorum_met() -> bool:
+            return ok_count[0] >= min(ctx.backward_k_min, len(live_experts))
+
+        pool = ThreadPoolExecutor(max_workers=min(32, max(1, len(live_experts))))
+        try:
             futures = {}
             for uid in live_experts:
                 pairs = ctx.expert_tokens[uid]
                 idx = [n for n, _ in pairs]
                 slots = [s for _, s in pairs]
+                idx_of[uid] = idx
                 grad_out = w[idx, slots].unsqueeze(1).to(g.dtype) * g[idx]
-                futures[uid] = (
-                    pool.submit(rpc_call, ctx.endpoint_of[uid], "bwd",
-                                [uid, [x_cpu[idx], grad_out]], ctx.backward_timeout),
-                    idx,
+                futures[uid] = pool.submit(
+                    rpc_call, ctx.endpoint_of[uid], "bwd",
+                    [uid, [x_cpu[idx], grad_out]], ctx.backward_timeout,
                 )
-            done, _ = futures_wait([f for f, _ in futures.values()], timeout=ctx.backward_timeout)
-            for uid, (fut, idx) in futures.items():
-                if fut in done and fut.exception() is None:
-                    grad_x_part = fut.result()[0].float()
-                    dx[idx] += grad_x_part
-                else:
-                    fut.cancel()
-                    logger.warning(f"backward RPC to expert {uid} failed/timed out; zero grads")
+            _wait_quorum_early_exit(futures, _on_success, _quorum_met,
+                                    ctx.backward_timeout, ctx.timeout_after_k_min, "backward")
+        finally:
+            pool.shutdown(wait=False)
 
-        ok = sum(1 for uid, (fut, _) in futures.items()
-                 if fut in done and fut.exception() is None)
-        if ok < min(ctx.backward_k_min, len(live_experts)):
+        if ok_count[0] < min(ctx.backward_k_min, len(live_experts)):
             # reference semantics: proceed anyway (failed experts contribute
             # zero grads), but surface that the backward quorum was missed
             logger.warning(
-                f"backward quorum missed: {ok}/{len(live_experts)} experts "
+                f"backward quorum missed: {ok_count[0]}/{len(live_experts)} experts "
                 f"returned grads (backward_k_min={ctx.backward_k_min})"
             )
 
         return (dx.to(ctx.device, dtype=grad_y.dtype),
-                dlogits.to(ctx.device), None, None, None, None, None, None)
+                dlogits.to(ctx.device), None, None, None, None, None, None, None)
