@@ -1,13 +1,25 @@
 """DecentralizedAverager — group parameter averaging with DHT matchmaking
 (SURVEY.md §2 "DecentralizedAverager", §3.4 call stack; Moshpit SGD rounds).
 
-Matchmaking: peers race to become the leader under the DHT key
-"{prefix}.leader" (liveness-leased). The leader accepts TCP 'join' requests
-until `averaging_expiration` elapses (or the group is full), then broadcasts
-the member list. One averaging round = butterfly all-reduce by parts:
+Matchmaking (Moshpit grid): the prefix owns `num_group_slots` concurrent
+group slots, each with its own liveness-leased leader key
+"{prefix}.slot{i}.leader". A peer scans the slots in random order and joins
+the first joinable leader ('join' TCP request); if every advertised leader
+is full/stale it becomes the leader of a random free slot. Leaders accept
+joiners until `averaging_expiration` elapses (or the group is full), then
+broadcast the member list. With P > target_group_size peers, multiple slots
+host groups CONCURRENTLY, and the random slot assignment re-shuffles group
+composition every round, so repeated rounds mix values globally (Moshpit
+SGD, arXiv:2103.03239) instead of serializing through one leader.
+
+One averaging round = butterfly all-reduce by parts:
 member i owns part i of the flat parameter vector; everyone sends part i to
 member i ('reduce' RPC), the owner returns the weighted average in the RPC
 reply (reduce-scatter + all-gather in one round trip per part).
+
+Each live averager also re-announces itself under a hashed per-peer state
+key "{prefix}.peer{h}" (liveness-leased), so `load_state_from_peers` can
+find ANY live peer for a cold start — not just a transient round leader.
 
 Fault tolerance: any member failing mid-round aborts this peer's round
 gracefully (tensors left unchanged; step() returns None) — the next round
@@ -97,6 +109,8 @@ class DecentralizedAverager:
         request_timeout: float = 5.0,
         listen_on: Endpoint = ("127.0.0.1", 0),
         compression: str = "none",
+        num_group_slots: int = 4,
+        state_announce_period: float = 2.0,
         start: bool = False,
     ):
         self.tensors = list(averaged_tensors)
@@ -111,6 +125,9 @@ class DecentralizedAverager:
         # reference's tensor-compression knob. All group peers must agree.
         assert compression in ("none", "bf16"), compression
         self.compression = compression
+        assert num_group_slots >= 1
+        self.num_group_slots = num_group_slots
+        self.state_announce_period = state_announce_period
         self._tensor_lock = threading.Lock()
         self._rounds: Dict[str, _RoundState] = {}
         self._rounds_lock = threading.Lock()
@@ -131,6 +148,20 @@ class DecentralizedAverager:
         self._server_sock.listen(64)
         self.port = self._server_sock.getsockname()[1]
         threading.Thread(target=self._accept_loop, daemon=True, name="averager_server").start()
+        threading.Thread(target=self._announce_loop, daemon=True, name="averager_announce").start()
+
+    def _announce_loop(self) -> None:
+        """Liveness-leased self-announcement so cold-starting peers can find
+        ANY live averager's state (don't rely on the transient
+        leader key). Lease = 2× period, matching the server announcer."""
+        key = self._state_key
+        while not self._stop.is_set():
+            try:
+                self.dht.store(key, list(self.endpoint),
+                               get_dht_time() + 2 * self.state_announce_period)
+            except Exception as e:  # noqa: BLE001
+                logger.debug(f"averager state announce failed: {e!r}")
+            self._stop.wait(self.state_announce_period)
 
     @property
     def endpoint(self) -> Endpoint:
@@ -272,28 +303,49 @@ class DecentralizedAverager:
             return None
         return PytorchSerializer.loads(payload)
 
+    _STATE_KEY_SLOTS = 16
+
+    def _slot_leader_key(self, slot: int) -> str:
+        return f"{self.prefix}.slot{slot}.leader"
+
     @property
-    def _leader_key(self) -> str:
-        return f"{self.prefix}.leader"
+    def _state_key(self) -> str:
+        import zlib
+
+        h = zlib.crc32(repr(self.endpoint).encode()) % self._STATE_KEY_SLOTS
+        return f"{self.prefix}.peer{h}"
 
     def step(self, weight: float = 1.0, timeout: Optional[float] = None) -> Optional[_GroupInfo]:
-        """One averaging round: matchmake, butterfly all-reduce, write back.
-        Returns the group info on success, None if no group formed / aborted."""
+        """One averaging round: matchmake on the slot grid, butterfly
+        all-reduce, write back. Returns the group info on success, None if no
+        group formed / aborted. Slot order is randomized per attempt so group
+        composition re-shuffles every round (Moshpit global mixing)."""
         deadline = time.monotonic() + (timeout if timeout is not None else
                                        3 * (self.averaging_expiration + self.request_timeout))
         while time.monotonic() < deadline and not self._stop.is_set():
-            leader_ep, expiration = self.dht.get(self._leader_key)
-            if leader_ep is not None and tuple(leader_ep) != self.endpoint:
+            slots = list(range(self.num_group_slots))
+            random.shuffle(slots)
+            free_slots = []
+            joined = None
+            for slot in slots:
+                leader_ep, _expiration = self.dht.get(self._slot_leader_key(slot))
+                if leader_ep is None or tuple(leader_ep) == self.endpoint:
+                    free_slots.append(slot)
+                    continue
                 reply = None
                 try:
                     reply = self._rpc(tuple(leader_ep), "join", [list(self.endpoint), weight])
                 except (OSError, RuntimeError):
-                    pass
+                    free_slots.append(slot)  # stale/dead leader: slot is takeable
+                    continue
                 if reply is not None:
-                    group = _GroupInfo(reply[0], [tuple(m) for m in reply[1]], reply[2])
-                    return self._run_allreduce(group, weight)
-                # leader busy/stale: fall through and try to lead
-            if self._try_lead(weight):
+                    joined = _GroupInfo(reply[0], [tuple(m) for m in reply[1]], reply[2])
+                    break
+                # leader full/closed: try the next slot
+            if joined is not None:
+                return self._run_allreduce(joined, weight)
+            lead_slot = random.choice(free_slots) if free_slots else random.choice(slots)
+            if self._try_lead(lead_slot, weight):
                 group = self._leader_state["final"]
                 self._leader_state = None
                 if group is not None:
@@ -304,8 +356,9 @@ class DecentralizedAverager:
             time.sleep(0.05 + random.random() * 0.2)
         return None
 
-    def _try_lead(self, weight: float) -> bool:
-        """Become leader: advertise in DHT, collect joiners until expiration."""
+    def _try_lead(self, slot: int, weight: float) -> bool:
+        """Become leader of `slot`: advertise in DHT, collect joiners until
+        expiration."""
         with self._leader_lock:
             group_id = os.urandom(8).hex()
             self._leader_state = {
@@ -316,7 +369,7 @@ class DecentralizedAverager:
                 "final": None,
                 "event": threading.Event(),
             }
-        stored = self.dht.store(self._leader_key, list(self.endpoint),
+        stored = self.dht.store(self._slot_leader_key(slot), list(self.endpoint),
                                 get_dht_time() + self.averaging_expiration)
         if not stored:
             with self._leader_lock:
@@ -386,18 +439,37 @@ class DecentralizedAverager:
                 self._rounds.pop(group.group_id, None)
 
     # ---- state sharing (cold start) ------------------------------------
+    def _live_peer_candidates(self) -> List[Endpoint]:
+        """All endpoints advertised under the prefix: per-peer state keys
+        (liveness-leased self-announcements) plus any current slot leaders."""
+        keys = [f"{self.prefix}.peer{h}" for h in range(self._STATE_KEY_SLOTS)]
+        keys += [self._slot_leader_key(s) for s in range(self.num_group_slots)]
+        seen, candidates = set(), []
+        for key in keys:
+            ep, _ = self.dht.get(key)
+            if ep is None:
+                continue
+            ep = tuple(ep)
+            if ep != self.endpoint and ep not in seen:
+                seen.add(ep)
+                candidates.append(ep)
+        return candidates
+
     def load_state_from_peers(self, timeout: Optional[float] = None) -> bool:
-        """Fetch current tensors from any live peer advertising under prefix."""
-        leader_ep, _ = self.dht.get(self._leader_key)
-        if leader_ep is None or tuple(leader_ep) == self.endpoint:
-            return False
-        try:
-            tensors = self._rpc(tuple(leader_ep), "get_state", None, timeout)
-        except (OSError, RuntimeError):
-            return False
-        if tensors is None:
-            return False
-        with self.get_tensors() as mine:
-            for dst, src in zip(mine, tensors):
-                dst.copy_(src.to(dst.dtype))
-        return True
+        """Fetch current tensors from ANY live peer advertised under the
+        prefix (per-peer state announcements, falling back to slot leaders) —
+        not just a transient round leader (ADVICE r1 low #5)."""
+        candidates = self._live_peer_candidates()
+        random.shuffle(candidates)
+        for ep in candidates:
+            try:
+                tensors = self._rpc(ep, "get_state", None, timeout)
+            except (OSError, RuntimeError):
+                continue
+            if tensors is None:
+                continue
+            with self.get_tensors() as mine:
+                for dst, src in zip(mine, tensors):
+                    dst.copy_(src.to(dst.dtype))
+            return True
+        return False
