@@ -102,7 +102,11 @@ class ExpertStack(nn.Module):
             self.b2_f32[idx].copy_(self.W2_bias.data[idx].to(self.b2_f32.device))
 
     def save_checkpoints(self, checkpoint_dir, uid_prefix: str, first_global_idx: int = 0,
-                         update_count: int = 0) -> None:
+                         update_count: int = 0, optimizer=None) -> None:
+        """Per-uid checkpoint files in the server/checkpoints.py on-disk
+        format. Pass the shard's optimizer (FusedExpertSGD / ExpertAdam) to
+        persist momentum / Adam moments / the SR step counter — a restart
+        is then a true RESUME, not just a weight reload (VERDICT r1 #5)."""
         import time as _time
         from pathlib import Path
 
@@ -111,7 +115,8 @@ class ExpertStack(nn.Module):
             uid = f"{uid_prefix}.{first_global_idx + i}"
             d = Path(checkpoint_dir) / uid
             d.mkdir(parents=True, exist_ok=True)
-            blob = {"model": self.export_expert_state(i), "optimizer": {},
+            opt_state = optimizer.state_for_expert(i) if optimizer is not None else {}
+            blob = {"model": self.export_expert_state(i), "optimizer": opt_state,
                     "update_count": update_count}
             tmp = d / f".tmp_{ts}.pt"
             torch.save(blob, tmp)
@@ -122,7 +127,8 @@ class ExpertStack(nn.Module):
 
             _shutil.copyfile(d / f"checkpoint_{ts}.pt", d / "checkpoint_last.pt")
 
-    def load_checkpoints(self, checkpoint_dir, uid_prefix: str, first_global_idx: int = 0) -> int:
+    def load_checkpoints(self, checkpoint_dir, uid_prefix: str, first_global_idx: int = 0,
+                         optimizer=None) -> int:
         from pathlib import Path
 
         n = 0
@@ -131,6 +137,8 @@ class ExpertStack(nn.Module):
             if f.exists():
                 blob = torch.load(f, map_location="cpu", weights_only=False)
                 self.import_expert_state(i, blob["model"])
+                if optimizer is not None and blob.get("optimizer"):
+                    optimizer.load_expert_state(i, blob["optimizer"])
                 n += 1
         return n
 
@@ -209,6 +217,27 @@ class FusedExpertSGD:
         st.zero_main_grads()
         self.step_count += 1
 
+    # ---- per-expert optimizer checkpoint slices ------------------------
+    def state_for_expert(self, idx: int) -> dict:
+        """Momentum slice + step counter of local expert `idx` (the SR seed
+        is derived from step_count, so persisting it keeps the stochastic-
+        rounding stream deterministic across a restart)."""
+        sd = {"type": "fused_sgd", "step_count": self.step_count}
+        if self.momentum_bufs is not None:
+            names = ("W1", "W2", "b1", "b2")
+            sd["momentum"] = {n: buf[idx].detach().cpu().clone()
+                              for n, buf in zip(names, self.momentum_bufs)}
+        return sd
+
+    @torch.no_grad()
+    def load_expert_state(self, idx: int, sd: dict) -> None:
+        self.step_count = max(self.step_count, int(sd.get("step_count", 0)))
+        if self.momentum_bufs is not None and sd.get("momentum"):
+            names = ("W1", "W2", "b1", "b2")
+            for n, buf in zip(names, self.momentum_bufs):
+                if n in sd["momentum"]:
+                    buf[idx].copy_(sd["momentum"][n].to(buf.device))
+
 
 @torch.no_grad()
 def adam_update_(p: torch.Tensor, g: torch.Tensor, m: torch.Tensor, v: torch.Tensor,
@@ -233,7 +262,9 @@ class ExpertAdam:
     Torch-level (hipBLASLt-free elementwise ops) on the fp32 masters, then
     re-quantizes the bf16 compute copies and their transposes — the default
     remains FusedExpertSGD, which fuses all of that into one kernel pass.
-    Requires init_gpu_state(master_weights=True) on GPU."""
+    GPU mode requires init_gpu_state(master_weights=True); on CPU (gloo
+    tests, resume tests) it reads autograd .grad of the masters instead of
+    the main_grad buffers and skips the compute-copy refresh."""
 
     def __init__(self, stack: ExpertStack, lr: float = 1e-3,
                  betas: Tuple[float, float] = (0.9, 0.999), eps: float = 1e-8,
@@ -250,32 +281,67 @@ class ExpertAdam:
             self._state[name] = (torch.zeros_like(like), torch.zeros_like(like))
         return self._state[name]
 
+    def _named_params_and_grads(self):
+        st = self.stack
+        if st._gpu_state_ready:
+            return (("W1", st.W1.data, st.main_grad_W1),
+                    ("W2", st.W2.data, st.main_grad_W2),
+                    ("b1", st.W1_bias.data, st.main_grad_b1),
+                    ("b2", st.W2_bias.data, st.main_grad_b2))
+        return (("W1", st.W1.data, st.W1.grad),
+                ("W2", st.W2.data, st.W2.grad),
+                ("b1", st.W1_bias.data, st.W1_bias.grad),
+                ("b2", st.W2_bias.data, st.W2_bias.grad))
+
     @torch.no_grad()
     def step(self) -> None:
         st = self.stack
-        if not st.grad_ready:
+        if st._gpu_state_ready and not st.grad_ready:
+            return
+        named = self._named_params_and_grads()
+        if any(g is None for _, _, g in named):
             return
         self.step_count += 1
         b1, b2 = self.betas
-        for name, p, g in (("W1", st.W1.data, st.main_grad_W1),
-                           ("W2", st.W2.data, st.main_grad_W2),
-                           ("b1", st.W1_bias.data, st.main_grad_b1),
-                           ("b2", st.W2_bias.data, st.main_grad_b2)):
+        for name, p, g in named:
             m, v = self._mv(name, p)
-            adam_update_(p, g, m, v, self.step_count, self.lr, b1, b2,
+            adam_update_(p, g.float(), m, v, self.step_count, self.lr, b1, b2,
                          self.eps, self.weight_decay)
-        # refresh bf16 compute copies + transposes + fp32 bias mirrors
-        st.W1c.copy_(st.W1.data.bfloat16())
-        st.W2c.copy_(st.W2.data.bfloat16())
-        if st.W1c.is_cuda and HAS_HIP:
-            st.W1t.copy_(_C.transpose_bf16(st.W1c))
-            st.W2t.copy_(_C.transpose_bf16(st.W2c))
+        if st._gpu_state_ready:
+            # refresh bf16 compute copies + transposes + fp32 bias mirrors
+            st.W1c.copy_(st.W1.data.bfloat16())
+            st.W2c.copy_(st.W2.data.bfloat16())
+            if st.W1c.is_cuda and HAS_HIP:
+                st.W1t.copy_(_C.transpose_bf16(st.W1c))
+                st.W2t.copy_(_C.transpose_bf16(st.W2c))
+            else:
+                st.W1t.copy_(st.W1c.transpose(1, 2).contiguous())
+                st.W2t.copy_(st.W2c.transpose(1, 2).contiguous())
+            st.b1_f32.copy_(st.W1_bias.data)
+            st.b2_f32.copy_(st.W2_bias.data)
+            st.zero_main_grads()
         else:
-            st.W1t.copy_(st.W1c.transpose(1, 2).contiguous())
-            st.W2t.copy_(st.W2c.transpose(1, 2).contiguous())
-        st.b1_f32.copy_(st.W1_bias.data)
-        st.b2_f32.copy_(st.W2_bias.data)
-        st.zero_main_grads()
+            for _, p_named in (("W1", st.W1), ("W2", st.W2),
+                               ("b1", st.W1_bias), ("b2", st.W2_bias)):
+                p_named.grad = None
+
+    # ---- per-expert optimizer checkpoint slices ------------------------
+    def state_for_expert(self, idx: int) -> dict:
+        sd = {"type": "adam", "step_count": self.step_count, "state": {}}
+        for name, (m, v) in self._state.items():
+            sd["state"][name] = {"m": m[idx].detach().cpu().clone(),
+                                 "v": v[idx].detach().cpu().clone()}
+        return sd
+
+    @torch.no_grad()
+    def load_expert_state(self, idx: int, sd: dict) -> None:
+        self.step_count = max(self.step_count, int(sd.get("step_count", 0)))
+        likes = {"W1": self.stack.W1.data, "W2": self.stack.W2.data,
+                 "b1": self.stack.W1_bias.data, "b2": self.stack.W2_bias.data}
+        for name, mv in sd.get("state", {}).items():
+            m, v = self._mv(name, likes[name])
+            m[idx].copy_(mv["m"].to(m.device))
+            v[idx].copy_(mv["v"].to(v.device))
 
 
 class _AllToAll(torch.autograd.Function):
