@@ -257,6 +257,26 @@ def main():
     ms_per_step = elapsed / args.steps * 1000.0
     total_tokens_per_s = tokens_per_step_per_rank * world_size * args.steps / elapsed
 
+    # truthful labels: only claim dp/averager when one ran
+    par = f"ep{world_size}"
+    if averager is not None:
+        par += f"+dp{world_size} (async SGD, RCCL averager every {args.averaging_period})"
+    else:
+        par += " (async expert SGD)"
+
+    # capacity-drop observability: fraction of routed
+    # (token, expert) assignments dropped by the capacity clamp, last step
+    drop_fracs = []
+    for block in model.blocks:
+        c = block.moe.last_counts
+        if c is not None:
+            cap = block.moe.capacity(tokens_per_step_per_rank)
+            total = float(c.sum().item())
+            kept = float(c.clamp(max=cap).sum().item())
+            if total > 0:
+                drop_fracs.append((total - kept) / total)
+    dropped_frac = round(sum(drop_fracs) / len(drop_fracs), 4) if drop_fracs else None
+
     if rank == 0:
         print(json.dumps({
             "metric": "training tokens/sec, Transformer-DMoE 256-expert",
@@ -276,7 +296,9 @@ def main():
                          f"E{cfg.num_experts} top{cfg.k_best} ffn{cfg.ffn_dim}",
                 "global_batch": B * world_size,
                 "seq_len": cfg.seq_len,
-                "parallelism": f"ep{world_size}+dp{world_size} (async SGD, Moshpit averager)",
+                "parallelism": par,
+                "capacity_factor": args.capacity_factor,
+                "dropped_token_frac": dropped_frac,
                 "last_loss": round(float(last_loss), 4) if last_loss is not None else None,
             },
         }))
