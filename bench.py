@@ -27,11 +27,16 @@ from learning_at_home_amd.optim import MasterSGD
 
 
 def build_model(cfg: TransformerDMoEConfig, args, device, group=None):
+    world = dist.get_world_size() if dist.is_initialized() else 1
+    # "auto": overlap the a2a with expert compute at N>1 (4 chunks), no
+    # chunking at N=1 where there is no collective to hide
+    chunks = (4 if world > 1 else 1) if args.a2a_chunks == 0 else args.a2a_chunks
+
     def moe_factory(layer_idx, c):
         return ExpertParallelMoE(
             d_model=c.d_model, ffn_dim=c.ffn_dim, num_experts=c.num_experts,
             k=c.k_best, capacity_factor=args.capacity_factor,
-            aux_loss_coef=args.aux_coef, group=group,
+            aux_loss_coef=args.aux_coef, group=group, a2a_chunks=chunks,
         )
 
     model = TransformerDMoE(cfg, moe_factory)

@@ -331,6 +331,42 @@ class TestMoEOnGPU:
         y2 = moe(x.detach())
         assert torch.isfinite(y2.float()).all()
 
+    def test_chunked_pipeline_matches_unchunked_gpu(self):
+        """Chunked a2a/compute pipeline at W=1 on GPU: exercises the side
+        HIP stream orchestration, the ExpertStack slice views through the
+        HIP kernels, and the per-range grad_ready tracking (the stale-
+        accumulate pitfall is GPU-only: TN kernels write vs accumulate)."""
+        torch.manual_seed(21)
+        moe_a = ExpertParallelMoE(d_model=256, ffn_dim=512, num_experts=8, k=2,
+                                  capacity_factor=4.0, router_noise_std=0.0,
+                                  aux_loss_coef=0.0, a2a_chunks=1).to(DEV)
+        torch.manual_seed(21)
+        moe_b = ExpertParallelMoE(d_model=256, ffn_dim=512, num_experts=8, k=2,
+                                  capacity_factor=4.0, router_noise_std=0.0,
+                                  aux_loss_coef=0.0, a2a_chunks=4).to(DEV)
+        moe_a.experts.init_gpu_state()
+        moe_b.experts.init_gpu_state()
+        moe_a.eval(); moe_b.eval()
+        for step in range(2):  # second step catches stale-grad accumulation
+            x = torch.randn(4, 64, 256, device=DEV, requires_grad=True)
+            x2 = x.detach().clone().requires_grad_(True)
+            y1, y2 = moe_a(x), moe_b(x2)
+            torch.cuda.synchronize()
+            assert torch.equal(y1, y2), (
+                f"step {step}: chunked fwd differs, max "
+                f"{(y1 - y2).abs().max().item()}")
+            y1.float().pow(2).mean().backward()
+            y2.float().pow(2).mean().backward()
+            torch.cuda.synchronize()
+            assert torch.equal(x.grad, x2.grad), f"step {step}: dx differs"
+            for n in ("main_grad_W1", "main_grad_W2", "main_grad_b1", "main_grad_b2"):
+                g1, g2 = getattr(moe_a.experts, n), getattr(moe_b.experts, n)
+                assert torch.equal(g1, g2), (
+                    f"step {step}: {n} differs, max {(g1 - g2).abs().max().item()}")
+            assert moe_a.experts.grad_ready and moe_b.experts.grad_ready
+            moe_a.experts.zero_main_grads()
+            moe_b.experts.zero_main_grads()
+
     def test_moe_matches_cpu_semantics(self):
         """Same weights + tokens: GPU fused path ≈ CPU reference path."""
         torch.manual_seed(12)
