@@ -74,6 +74,9 @@ def main():
     p.add_argument("--batch", type=int, default=96, help="per-GPU batch size")
     p.add_argument("--vocab", type=int, default=32000)
     p.add_argument("--capacity-factor", type=float, default=1.5)
+    p.add_argument("--a2a-chunks", type=int, default=0,
+                   help="0 = auto (4 at N>1 to overlap the dispatch/return "
+                        "all-to-all with expert compute, 1 at N=1)")
     p.add_argument("--aux-coef", type=float, default=0.01)
     p.add_argument("--lr", type=float, default=0.01)
     p.add_argument("--averaging-period", type=int, default=10)
