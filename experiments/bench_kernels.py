@@ -57,8 +57,38 @@ def bench_sgd_t(E, R, C, tag):
     print(f"sgd_t[{tag}] {n/1e6:.0f}M params: {dt*1e3:7.3f} ms  {gbps:7.0f} GB/s eff")
 
 
+def bench_attention(B, H, S, tag):
+    """Flash attention vs torch sdpa at bench shapes, fwd and fwd+bwd.
+    Causal flops: QK^T + PV = 2 * 2 * BH * (S^2/2) * D per fwd pass."""
+    import torch.nn.functional as F
+
+    from learning_at_home_amd.ops.functional import flash_attention
+
+    D = 64
+    flops_fwd = 2.0 * 2 * B * H * (S * S / 2) * D
+    q, k, v = (torch.randn(B, H, S, D, device=DEV, dtype=torch.bfloat16) for _ in range(3))
+
+    dt = timeit(lambda: _C.flash_attn_fwd(q, k, v), iters=30)
+    print(f"attn_flash_fwd[{tag}]: {dt*1e3:7.3f} ms  {flops_fwd/dt/1e12:7.1f} TF")
+    dt = timeit(lambda: F.scaled_dot_product_attention(q, k, v, is_causal=True), iters=30)
+    print(f"attn_sdpa_fwd [{tag}]: {dt*1e3:7.3f} ms  {flops_fwd/dt/1e12:7.1f} TF")
+
+    def fb(fn):
+        qkv = [t.detach().clone().requires_grad_(True) for t in (q, k, v)]
+        out = fn(*qkv)
+        out.sum().backward()
+
+    flops_fb = flops_fwd * 3.5  # bwd ~ 2.5x fwd (dq, dkv recompute)
+    dt = timeit(lambda: fb(flash_attention), iters=15)
+    print(f"attn_flash_f+b[{tag}]: {dt*1e3:7.3f} ms  {flops_fb/dt/1e12:7.1f} TF")
+    dt = timeit(lambda: fb(lambda a, b, c: F.scaled_dot_product_attention(a, b, c, is_causal=True)), iters=15)
+    print(f"attn_sdpa_f+b [{tag}]: {dt*1e3:7.3f} ms  {flops_fb/dt/1e12:7.1f} TF")
+
+
 def main():
     torch.manual_seed(0)
+    print("=== attention (flagship: B=96 H=16 S=1024 D=64) ===")
+    bench_attention(96, 16, 1024, "flagship")
     print("=== flagship shapes (E=256, C=192, d=1024, ffn=4096) ===")
     bench_gemm_nt(256, 192, 4096, 1024, 2, "fwd1")
     bench_gemm_nt(256, 192, 1024, 4096, 1, "fwd2")

@@ -246,7 +246,7 @@ def flash_attention_fwd(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> to
     online softmax, MFMA QK^T and PV, one wave per 16 query rows); elsewhere
     it falls back to torch sdpa with identical causal semantics."""
     if q.is_cuda and HAS_HIP and q.dtype == torch.bfloat16 and q.shape[-1] == 64 \
-            and q.shape[-2] % 64 == 0:
+            and q.shape[-2] % 128 == 0:
         return _C.flash_attn_fwd(q.contiguous(), k.contiguous(), v.contiguous())
     return F.scaled_dot_product_attention(q, k, v, is_causal=True)
 
@@ -278,7 +278,7 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor) -> torch.
     """Differentiable causal attention. CDNA4 flash kernels when the shapes
     qualify ([B, H, S, 64] bf16 on GPU, S % 64 == 0), torch sdpa otherwise."""
     if q.is_cuda and HAS_HIP and q.dtype == torch.bfloat16 and q.shape[-1] == 64 \
-            and q.shape[-2] % 64 == 0:
+            and q.shape[-2] % 128 == 0:
         return _FlashAttention.apply(q, k, v)
     return F.scaled_dot_product_attention(q, k, v, is_causal=True)
 

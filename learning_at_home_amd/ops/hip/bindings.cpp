@@ -10,10 +10,11 @@ void lah_batched_gemm_nt(const void*, const void*, void*, const void*, void*, in
                          int, int, int, hipStream_t);
 void lah_batched_gemm_tn(const void*, const void*, void*, int, int, int, int, int,
                          hipStream_t);
-void lah_flash_attn_fwd(const void*, const void*, const void*, void*, void*, int, int,
+void lah_flash_attn_fwd(const void*, const void*, const void*, void*, void*, void*, int, int,
                         hipStream_t);
 void lah_flash_attn_bwd(const void*, const void*, const void*, const void*, const void*,
-                        const void*, void*, void*, void*, int, int, hipStream_t);
+                        const void*, void*, void*, void*, void*, void*, void*, int, int,
+                        hipStream_t);
 void lah_gating_topk_fwd(const void*, void*, void*, void*, int, int, int, hipStream_t);
 void lah_gating_topk_bwd(const void*, const void*, const void*, const void*, const void*,
                          void*, int, int, int, hipStream_t);
@@ -233,14 +234,16 @@ static void check_attn_args(const torch::Tensor& q, const torch::Tensor& k,
   CHECK_BF16(k); CHECK_CONTIG(k); CHECK_BF16(v); CHECK_CONTIG(v);
   TORCH_CHECK(q.dim() == 4 && q.size(3) == 64, "q must be [B, H, S, 64]");
   TORCH_CHECK(q.sizes() == k.sizes() && q.sizes() == v.sizes(), "q/k/v shape mismatch");
-  TORCH_CHECK(q.size(2) % 64 == 0, "seq_len must be a multiple of 64");
+  TORCH_CHECK(q.size(2) % 128 == 0, "seq_len must be a multiple of 128");
 }
 
 torch::Tensor flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v) {
   check_attn_args(q, k, v);
   auto o = torch::empty_like(q);
-  lah_flash_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(), nullptr,
-                     q.size(0) * q.size(1), q.size(2), cur_stream());
+  auto vt = torch::empty_like(v);  // [BH, 64, S] scratch via torch allocator
+  lah_flash_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), vt.data_ptr(),
+                     o.data_ptr(), nullptr, q.size(0) * q.size(1), q.size(2),
+                     cur_stream());
   return o;
 }
 
@@ -249,10 +252,12 @@ std::vector<torch::Tensor> flash_attn_fwd_lse(torch::Tensor q, torch::Tensor k,
                                               torch::Tensor v) {
   check_attn_args(q, k, v);
   auto o = torch::empty_like(q);
+  auto vt = torch::empty_like(v);
   auto lse = torch::empty({q.size(0), q.size(1), q.size(2)},
                           q.options().dtype(at::kFloat));
-  lah_flash_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(),
-                     lse.data_ptr(), q.size(0) * q.size(1), q.size(2), cur_stream());
+  lah_flash_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), vt.data_ptr(),
+                     o.data_ptr(), lse.data_ptr(), q.size(0) * q.size(1), q.size(2),
+                     cur_stream());
   return {o, lse};
 }
 
@@ -265,8 +270,12 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor q, torch::Tensor k,
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
+  auto qt = torch::empty_like(q);   // transposed-operand scratch
+  auto kt = torch::empty_like(k);
+  auto dot = torch::empty_like(dout);
   lah_flash_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), dout.data_ptr(),
-                     lse.data_ptr(), d.data_ptr(), dq.data_ptr(), dk.data_ptr(),
+                     lse.data_ptr(), d.data_ptr(), qt.data_ptr(), kt.data_ptr(),
+                     dot.data_ptr(), dq.data_ptr(), dk.data_ptr(),
                      dv.data_ptr(), q.size(0) * q.size(1), q.size(2), cur_stream());
   return {dq, dk, dv};
 }

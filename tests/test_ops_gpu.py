@@ -388,7 +388,7 @@ def test_flash_attention_fwd_vs_sdpa():
     from learning_at_home_amd.ops.functional import flash_attention_fwd
 
     torch.manual_seed(0)
-    for B, H, S in ((2, 4, 256), (1, 2, 64), (1, 1, 512)):
+    for B, H, S in ((2, 4, 256), (1, 2, 128), (1, 1, 512)):
         q = torch.randn(B, H, S, 64, device="cuda:0", dtype=torch.bfloat16)
         k = torch.randn(B, H, S, 64, device="cuda:0", dtype=torch.bfloat16)
         v = torch.randn(B, H, S, 64, device="cuda:0", dtype=torch.bfloat16)
