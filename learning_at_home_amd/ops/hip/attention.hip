@@ -34,6 +34,12 @@
 #define LN2 0.6931471805599453f
 #define NEG_BIG (-3.0e38f)
 
+// raw v_exp_f32 / v_log_f32 / v_rcp_f32: softmax arguments are <= 0 and
+// l >= 1, so libm's denorm/range fixup code (~5 VALU per call) buys nothing
+LAH_DEVINL float fast_exp2(float x) { return __builtin_amdgcn_exp2f(x); }
+LAH_DEVINL float fast_log2(float x) { return __builtin_amdgcn_logf(x); }
+LAH_DEVINL float fast_rcp(float x) { return __builtin_amdgcn_rcpf(x); }
+
 // ---------------------------------------------------------------------------
 // forward: one wave = 32 q rows; block = 4 waves = 128 q rows of one (b,h)
 // ---------------------------------------------------------------------------
@@ -64,17 +70,21 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
       aq[m][kk] = *(const bf16x8*)(Q + base + (long)(qw + m * 16 + l15) * 64 + kk * 32 + lk8);
 
   f32x4 of[2][4];
-  float m_run[2][4], l_run[2][4];
+  f32x4 lf[2];  // softmax denominator, accumulated as the P @ [V | 1]
+                // ones-column fragment — no cross-lane sum reduction needed
+  float m_run[2][4];
 #pragma unroll
   for (int m = 0; m < 2; ++m) {
 #pragma unroll
     for (int f = 0; f < 4; ++f) of[m][f] = {0.f, 0.f, 0.f, 0.f};
+    lf[m] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      m_run[m][r] = NEG_BIG;
-      l_run[m][r] = 0.f;
-    }
+    for (int r = 0; r < 4; ++r) m_run[m][r] = NEG_BIG;
   }
+  // ones-column B fragment: column n=0 of a 16-wide block is 1, rest 0
+  bf16x8 bones;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) bones[j] = (l15 == 0) ? (__bf16)1.0f : (__bf16)0.0f;
 
   const int kb_last = (qw + 31) >> 6;  // this wave's causal bound
   for (int kb = 0; kb <= kb_last; ++kb) {
@@ -130,42 +140,33 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
 #pragma unroll
         for (int r = 0; r < 4; ++r)
           rmax[m][r] = fmaxf(rmax[m][r], __shfl_xor(rmax[m][r], x, 64));
-    // ---- online softmax (exp2) + P -> wave-private LDS + O rescale -------
-    float corr[2][4], rsum[2][4];
+    // ---- online softmax (raw v_exp) + P -> wave-private LDS + rescale ----
+    float corr[2][4];
 #pragma unroll
     for (int m = 0; m < 2; ++m)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         float m_new = fmaxf(m_run[m][r], rmax[m][r]);
-        corr[m][r] = exp2f(m_run[m][r] - m_new);
+        corr[m][r] = fast_exp2(m_run[m][r] - m_new);
         m_run[m][r] = m_new;
-        rsum[m][r] = 0.f;
       }
 #pragma unroll
     for (int m = 0; m < 2; ++m)
 #pragma unroll
       for (int f = 0; f < 4; ++f)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          float p = exp2f(sf[m][f][r] - m_run[m][r]);
-          rsum[m][r] += p;
-          Ps[(m * 16 + r0 + r) * AT_LD + f * 16 + l15] = f2bf(p);
-        }
-#pragma unroll
-    for (int x = 1; x < 16; x <<= 1)
-#pragma unroll
-      for (int m = 0; m < 2; ++m)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) rsum[m][r] += __shfl_xor(rsum[m][r], x, 64);
+        for (int r = 0; r < 4; ++r)
+          Ps[(m * 16 + r0 + r) * AT_LD + f * 16 + l15] =
+              f2bf(fast_exp2(sf[m][f][r] - m_run[m][r]));
 #pragma unroll
     for (int m = 0; m < 2; ++m)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        l_run[m][r] = l_run[m][r] * corr[m][r] + rsum[m][r];
+        lf[m][r] *= corr[m][r];
 #pragma unroll
         for (int f = 0; f < 4; ++f) of[m][f][r] *= corr[m][r];
       }
-    // ---- O += P @ V: A = P rows (LDS), B-frag = Vt rows (global) ---------
+    // ---- [O | l] += P @ [V | 1]: A = P rows (LDS), B = Vt rows (global) --
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       bf16x8 ap0 = *(const bf16x8*)(Ps + (0 * 16 + l15) * AT_LD + kk * 32 + lk8);
@@ -176,10 +177,20 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
         of[0][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap0, bv, of[0][f], 0, 0, 0);
         of[1][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap1, bv, of[1][f], 0, 0, 0);
       }
+      lf[0] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap0, bones, lf[0], 0, 0, 0);
+      lf[1] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap1, bones, lf[1], 0, 0, 0);
     }
   }
 
-  // ---- epilogue: O / l; LSE back to natural log -------------------------
+  // ---- epilogue: broadcast l from column 0, O * (1/l), LSE --------------
+  float l_run[2][4], inv_l[2][4];
+#pragma unroll
+  for (int m = 0; m < 2; ++m)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      l_run[m][r] = __shfl(lf[m][r], lane & 48, 64);  // col-0 lane of the group
+      inv_l[m][r] = fast_rcp(l_run[m][r]);
+    }
 #pragma unroll
   for (int m = 0; m < 2; ++m)
 #pragma unroll
@@ -187,7 +198,7 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = qw + m * 16 + r0 + r;
-        O[base + (long)row * 64 + f * 16 + l15] = f2bf(of[m][f][r] / l_run[m][r]);
+        O[base + (long)row * 64 + f * 16 + l15] = f2bf(of[m][f][r] * inv_l[m][r]);
       }
   if (LSE != nullptr && l15 == 0) {
 #pragma unroll
@@ -195,7 +206,7 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int row = qw + m * 16 + r0 + r;
-        LSE[bh * S + row] = (m_run[m][r] + log2f(l_run[m][r])) * LN2;
+        LSE[bh * S + row] = (m_run[m][r] + fast_log2(l_run[m][r])) * LN2;
       }
   }
 }
@@ -271,7 +282,7 @@ void flash_attn_bwd_dq_kernel(const __bf16* __restrict__ Q, const __bf16* __rest
         for (int r = 0; r < 4; ++r) {
           int qg = qw + m * 16 + r0 + r;
           int kg = kb * 64 + f * 16 + l15;
-          float p = (kg <= qg) ? exp2f(sf[m][f][r] * scale2 - lse2[m][r]) : 0.f;
+          float p = (kg <= qg) ? fast_exp2(sf[m][f][r] * scale2 - lse2[m][r]) : 0.f;
           float ds = p * (df[m][f][r] - d_r[m][r]) * scale;
           dSs[(m * 16 + r0 + r) * AT_LD + f * 16 + l15] = f2bf(ds);
         }
@@ -373,7 +384,7 @@ void flash_attn_bwd_dkv_kernel(const __bf16* __restrict__ Q, const __bf16* __res
           int kg = kw + m * 16 + r0 + r;
           int qg = q0 + f * 16 + l15;
           float p = (!boundary || kg <= qg)
-                        ? exp2f(stf[m][f][r] * scale2 - lse2_c[f]) : 0.f;
+                        ? fast_exp2(stf[m][f][r] * scale2 - lse2_c[f]) : 0.f;
           float ds = p * (dpf[m][f][r] - d_c[f]) * scale;
           int lrow = (m * 16 + r0 + r) * AT_LD + f * 16 + l15;
           Ps[lrow] = f2bf(p);

@@ -134,13 +134,49 @@ class ElasticGroup:
         alive = self.alive_slots()
         return sorted(alive) != sorted(self.members)
 
+    # ---- bounded-time comm teardown (SURVEY §7 hard part (b)) ----------
+    def abort_comm(self, timeout: float = 10.0) -> None:
+        """Tear down the current process group WITHOUT deadlocking on
+        in-flight collectives. A rank that died mid-all-to-all leaves
+        survivors' RCCL kernels spinning; plain destroy_process_group would
+        then block on the pending work. _abort_process_group kills the
+        communicator (and its in-flight ops) immediately; the watchdogged
+        destroy is the fallback for backends without an abort (gloo)."""
+        if not dist.is_initialized():
+            self.group = None
+            return
+        try:
+            from torch.distributed import distributed_c10d as c10d
+
+            if hasattr(c10d, "_abort_process_group"):
+                c10d._abort_process_group()  # aborts NCCL/RCCL comms + pending work
+        except Exception as e:  # noqa: BLE001
+            logger.warning(f"slot {self.slot}: process-group abort failed ({e!r})")
+        done = threading.Event()
+
+        def _destroy():
+            try:
+                if dist.is_initialized():
+                    dist.destroy_process_group()
+            except Exception as e:  # noqa: BLE001
+                logger.warning(f"slot {self.slot}: destroy after abort failed ({e!r})")
+            done.set()
+
+        t = threading.Thread(target=_destroy, daemon=True, name="pg_destroy")
+        t.start()
+        if not done.wait(timeout):
+            logger.warning(
+                f"slot {self.slot}: destroy_process_group stuck > {timeout}s "
+                "after abort; abandoning it (daemon thread) and rebuilding fresh"
+            )
+        self.group = None
+
     def form_group(self, expected: Optional[int] = None, deadline: float = 90.0) -> None:
         """(Re-)form the process group over the currently alive slots.
         Every alive rank must call this (coordinator publishes, others poll)."""
         t_end = time.monotonic() + deadline
         if dist.is_initialized():
-            dist.destroy_process_group()
-            self.group = None
+            self.abort_comm()
         while time.monotonic() < t_end and not self._stop.is_set():
             self.announce()  # our own lease may have lapsed during a stall
             alive = self._stable_alive()
@@ -177,8 +213,7 @@ class ElasticGroup:
                 return
             except Exception as e:  # noqa: BLE001
                 logger.warning(f"slot {self.slot}: gen {gen_no} join failed ({e!r}); retrying")
-                if dist.is_initialized():
-                    dist.destroy_process_group()
+                self.abort_comm()
                 # JITTERED backoff: identical retry periods phase-lock the
                 # whole group into a rendezvous livelock (each member is
                 # inside a doomed init exactly when the next gen appears —
@@ -198,8 +233,7 @@ class ElasticGroup:
         """Graceful departure: stop announcing; the lease lapses within
         `lease` seconds and survivors re-form without us."""
         self._stop.set()
-        if dist.is_initialized():
-            dist.destroy_process_group()
+        self.abort_comm()
 
     def shutdown(self) -> None:
         self.leave()

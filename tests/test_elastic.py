@@ -128,3 +128,61 @@ def test_drop_and_rejoin_rank():
             if p.is_alive():
                 p.terminate()
         seed.shutdown()
+
+
+def _abort_worker(rank, world, port):
+    """Rank 1 dies before joining rank 0's collective; rank 0's all_reduce
+    times out, abort_comm() must return within bounded time and allow a
+    fresh world-1 re-init (SURVEY §7 hard part (b): no deadlock on in-flight
+    collectives during teardown)."""
+    import datetime
+
+    from learning_at_home_amd.parallel.elastic import ElasticGroup
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world,
+                            timeout=datetime.timedelta(seconds=3))
+    if rank == 1:
+        return  # dies without participating
+    t = torch.ones(4)
+    raised = False
+    try:
+        dist.all_reduce(t)  # hangs until the 3 s gloo timeout, then raises
+    except Exception:
+        raised = True
+    assert raised, "collective with a dead peer should have timed out"
+    # bounded-time teardown despite the wedged collective
+    group = ElasticGroup.__new__(ElasticGroup)
+    group.slot = 0
+    group.group = None
+    import threading as _th
+
+    group._stop = _th.Event()
+    t0 = time.monotonic()
+    group.abort_comm(timeout=10.0)
+    assert time.monotonic() - t0 < 15.0, "abort_comm did not bound teardown time"
+    assert not dist.is_initialized(), "process group still up after abort"
+    # survivor re-forms alone and can use collectives again
+    dist.init_process_group("gloo", rank=0, world_size=1,
+                            init_method=f"tcp://127.0.0.1:{port + 1}")
+    t2 = torch.full((4,), 2.0)
+    dist.all_reduce(t2)
+    assert torch.equal(t2, torch.full((4,), 2.0))
+    dist.destroy_process_group()
+
+
+def test_abort_comm_recovers_from_hung_collective():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_abort_worker, args=(r, 2, port)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(90)
+    assert procs[0].exitcode == 0, "survivor rank failed abort/recovery"
+    assert procs[1].exitcode == 0
