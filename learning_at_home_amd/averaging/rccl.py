@@ -96,3 +96,24 @@ class RcclAverager:
         """Make the CURRENT stream wait until the averaged values are written."""
         if self.use_cuda and self._done_event is not None:
             torch.cuda.current_stream(self.device).wait_event(self._done_event)
+
+    @torch.no_grad()
+    def rebind(self, process_group: Optional[dist.ProcessGroup] = None) -> None:
+        """Adopt a REBUILT process group after an elastic membership change
+        (parallel/elastic.py generation rebuild): world size, padding and
+        shard sizing are re-derived for the new group. Callers invoke this
+        between rounds, after ElasticGroup.form_group() — any in-flight
+        round on the old communicator was already torn down by abort_comm,
+        and wait() here fences the side stream before buffers are resized."""
+        self.wait()
+        assert dist.is_initialized(), "rebind requires an initialized process group"
+        self.group = process_group
+        self.world_size = dist.get_world_size(process_group)
+        numel = sum(t.numel() for t in self.tensors)
+        padded = ((numel + self.world_size - 1) // self.world_size) * self.world_size
+        if padded != self.padded or self.flat.numel() != padded:
+            self.padded = padded
+            self.flat = torch.zeros(padded, dtype=self.dtype, device=self.device)
+        shard_n = padded // self.world_size
+        if self.shard.numel() != shard_n:
+            self.shard = torch.zeros(shard_n, dtype=self.dtype, device=self.device)

@@ -186,3 +186,52 @@ def test_abort_comm_recovers_from_hung_collective():
         p.join(90)
     assert procs[0].exitcode == 0, "survivor rank failed abort/recovery"
     assert procs[1].exitcode == 0
+
+
+def _rebind_worker(rank, world, port):
+    """RcclAverager survives an elastic generation rebuild: average at
+    world=2, lose a rank, abort, re-init at world=1, rebind, keep stepping."""
+    from learning_at_home_amd.averaging import RcclAverager
+    from learning_at_home_amd.parallel.elastic import ElasticGroup
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    params = [torch.full((6,), float(rank))]
+    av = RcclAverager(params, dtype=torch.float32)
+    av.step(async_op=False)
+    assert torch.allclose(params[0], torch.full((6,), 0.5)), params[0]
+    dist.barrier()  # both ranks finished round 1 before rank 1 departs
+    if rank == 1:
+        dist.destroy_process_group()
+        return  # rank 1 leaves the job
+    # survivor: generation rebuild to world=1, averager rebinds and still works
+    group = ElasticGroup.__new__(ElasticGroup)
+    group.slot = 0
+    group.group = None
+    import threading as _th
+
+    group._stop = _th.Event()
+    group.abort_comm(timeout=10.0)
+    dist.init_process_group("gloo", rank=0, world_size=1,
+                            init_method=f"tcp://127.0.0.1:{port + 1}")
+    av.rebind()
+    params[0].fill_(7.0)
+    av.step(async_op=False)  # world 1: identity, must not hang or resize-crash
+    assert torch.allclose(params[0], torch.full((6,), 7.0))
+    dist.destroy_process_group()
+
+
+def test_rccl_averager_rebind_after_membership_change():
+    import socket
+
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_rebind_worker, args=(r, 2, port)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(90)
+    assert all(p.exitcode == 0 for p in procs), [p.exitcode for p in procs]

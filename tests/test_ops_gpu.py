@@ -399,6 +399,29 @@ def test_flash_attention_fwd_vs_sdpa():
         assert err < 3e-2, f"B{B} H{H} S{S}: max err {err}"
 
 
+def test_flash_attention_defer_max_rescale_branch():
+    """The defer-max rescale branch is rare and data-dependent (guide §5.4
+    rule 26): force it with spiked K rows late in the sequence so the
+    running max jumps past DEFER_THR mid-loop, and check against fp32 sdpa."""
+    import torch.nn.functional as F
+
+    from learning_at_home_amd.ops.functional import flash_attention_fwd
+
+    torch.manual_seed(3)
+    B, H, S = 2, 2, 1024
+    q = torch.randn(B, H, S, 64, device="cuda:0", dtype=torch.bfloat16)
+    k = torch.randn(B, H, S, 64, device="cuda:0", dtype=torch.bfloat16)
+    v = torch.randn(B, H, S, 64, device="cuda:0", dtype=torch.bfloat16)
+    # spike several late K rows: raw q.k there ≫ every earlier score, so the
+    # deferred max must rescale O/l exactly once per affected row
+    for t in (700, 701, 900):
+        k[:, :, t, :] = q[:, :, t, :] * 40.0
+    out = flash_attention_fwd(q, k, v)
+    ref = F.scaled_dot_product_attention(q.float(), k.float(), v.float(), is_causal=True)
+    err = (out.float() - ref).abs().max().item()
+    assert err < 5e-2, f"spiked-K rescale path wrong: max err {err}"
+
+
 def test_flash_attention_bwd_vs_sdpa():
     """Differentiable flash attention: grads vs fp32 torch sdpa autograd."""
     import torch.nn.functional as F
