@@ -67,7 +67,8 @@ LAH_DEVINL void attn_block_id(int nqb, int BH, int& qb, long& bh) {
 // ---------------------------------------------------------------------------
 // forward: one wave = 32 q rows; block = 4 waves = 128 q rows of one (b,h)
 // ---------------------------------------------------------------------------
-__launch_bounds__(256, 2) __global__
+template <int MINW, bool PREFETCH>
+__launch_bounds__(256, MINW) __global__
 void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
                            const __bf16* __restrict__ K,   // [BH, S, 64]
                            const __bf16* __restrict__ Vt,  // [BH, 64, S]
@@ -221,15 +222,22 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
     }
   };
 
-  load_k(bkA, 0);
-  int kb = 0;
-  while (true) {  // 2-stage unroll: static buffer naming, no runtime index
-    if (kb + 1 <= kb_last) load_k(bkB, kb + 1);
-    tile(bkA, kb);
-    if (++kb > kb_last) break;
-    if (kb + 1 <= kb_last) load_k(bkA, kb + 1);
-    tile(bkB, kb);
-    if (++kb > kb_last) break;
+  if constexpr (PREFETCH) {
+    load_k(bkA, 0);
+    int kb = 0;
+    while (true) {  // 2-stage unroll: static buffer naming, no runtime index
+      if (kb + 1 <= kb_last) load_k(bkB, kb + 1);
+      tile(bkA, kb);
+      if (++kb > kb_last) break;
+      if (kb + 1 <= kb_last) load_k(bkA, kb + 1);
+      tile(bkB, kb);
+      if (++kb > kb_last) break;
+    }
+  } else {
+    for (int kb = 0; kb <= kb_last; ++kb) {
+      load_k(bkA, kb);
+      tile(bkA, kb);
+    }
   }
 
   // ---- epilogue: broadcast l from column 0, O * (1/l), LSE --------------
@@ -482,7 +490,16 @@ extern "C" void lah_flash_attn_fwd(const void* Q, const void* K, const void* V,
   lah_transpose_bf16(V, Vt_scratch, BH, S, 64, stream);
   dim3 grid((S / 128) * BH), block(256);
   const float scale2 = (1.0f / sqrtf(64.0f)) * LOG2E;
-  hipLaunchKernelGGL(flash_attn_fwd_kernel, grid, block, 0, stream,
+  // occupancy/prefetch variant A/B knob (perf work; default = measured best)
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("LAH_ATTN_V");
+    v = e ? atoi(e) : 0;
+  }
+  auto kern = flash_attn_fwd_kernel<2, true>;       // v0: 2 waves + K prefetch
+  if (v == 1) kern = flash_attn_fwd_kernel<3, false>;  // v1: 3 waves, no prefetch
+  if (v == 2) kern = flash_attn_fwd_kernel<3, true>;   // v2: 3 waves + prefetch (spills)
+  hipLaunchKernelGGL(kern, grid, block, 0, stream,
                      (const __bf16*)Q, (const __bf16*)K, (const __bf16*)Vt_scratch,
                      (__bf16*)O, (float*)LSE, BH, S, scale2);
 }

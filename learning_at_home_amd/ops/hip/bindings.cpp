@@ -88,7 +88,7 @@ void batched_gemm_tn(torch::Tensor A, torch::Tensor B, torch::Tensor C,
 std::vector<torch::Tensor> gating_topk_fwd(torch::Tensor logits, int64_t k) {
   CHECK_CUDA(logits); CHECK_F32(logits); CHECK_CONTIG(logits);
   int T = logits.size(0), E = logits.size(1);
-  TORCH_CHECK(E <= 1024, "gating kernel supports up to 1024 experts, got ", E);
+  TORCH_CHECK(E <= 4096, "gating kernel supports up to 4096 experts, got ", E);
   TORCH_CHECK(k == 1 || k == 2 || k == 4 || k == 8,
               "gating kernel supports k in {1,2,4,8}, got ", k);
   auto probs = torch::empty({T, E}, logits.options().dtype(at::kBFloat16));
@@ -104,7 +104,7 @@ torch::Tensor gating_topk_bwd(torch::Tensor probs, torch::Tensor topk_idx,
                               c10::optional<torch::Tensor> aux_g) {
   CHECK_BF16(probs); CHECK_I32(topk_idx); CHECK_F32(topk_w); CHECK_F32(d_topk_w);
   int T = probs.size(0), E = probs.size(1), k = topk_idx.size(1);
-  TORCH_CHECK(E <= 1024, "gating kernel supports up to 1024 experts, got ", E);
+  TORCH_CHECK(E <= 4096, "gating kernel supports up to 4096 experts, got ", E);
   auto d_logits = torch::empty({T, E}, topk_w.options());
   const void* aux_p = nullptr;
   if (aux_g.has_value()) { CHECK_F32(*aux_g); CHECK_CONTIG(*aux_g); aux_p = aux_g->data_ptr(); }

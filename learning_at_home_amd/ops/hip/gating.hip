@@ -179,10 +179,13 @@ extern "C" void lah_gating_topk_fwd(const void* logits, void* probs, void* topk_
     return;                                                                            \
   }
 #define CASE(N) CASE_K(N, 1) CASE_K(N, 2) CASE_K(N, 4) CASE_K(N, 8)
-  CASE(1) CASE(2) CASE(4) CASE(8) CASE(16)
+  // EPL 32/64 lift the wall to E <= 4096: 2*EPL VGPRs of
+  // per-lane logit state, so occupancy drops for the huge-E tail — priced
+  // only when such a config is actually run
+  CASE(1) CASE(2) CASE(4) CASE(8) CASE(16) CASE(32) CASE(64)
 #undef CASE
 #undef CASE_K
-  abort();  // E > 1024 or k not in {1,2,4,8}: unsupported
+  abort();  // E > 4096 or k not in {1,2,4,8}: unsupported (bindings TORCH_CHECK)
 }
 
 extern "C" void lah_gating_topk_bwd(const void* probs, const void* topk_idx,
@@ -199,7 +202,7 @@ extern "C" void lah_gating_topk_bwd(const void* probs, const void* topk_idx,
                        (const float*)aux_g, (float*)d_logits, T, E, k);                \
     return;                                                                            \
   }
-  CASE(1) CASE(2) CASE(4) CASE(8) CASE(16)
+  CASE(1) CASE(2) CASE(4) CASE(8) CASE(16) CASE(32) CASE(64)
 #undef CASE
   abort();
 }

@@ -94,7 +94,8 @@ class TestBatchedGemmTN:
 
 
 class TestGating:
-    @pytest.mark.parametrize("E,k", [(64, 4), (256, 4), (1024, 1), (256, 8)])
+    @pytest.mark.parametrize("E,k", [(64, 4), (256, 4), (1024, 1), (256, 8),
+                                     (2048, 4), (4096, 1)])
     def test_fwd(self, E, k):
         torch.manual_seed(4)
         T = 257
