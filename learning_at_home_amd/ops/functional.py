@@ -220,13 +220,30 @@ class _BatchedFFNHip(torch.autograd.Function):
         stack = ctx.stack
         dy = dy.contiguous()
         # dh_pre = (dy @ W2t^T) * der; dx = dh_pre @ W1t^T   (MFMA NT)
+        # (computed BEFORE any fused weight update below: dx must see the
+        # same weights the forward used)
         dh_pre = _C.batched_gemm_nt(dy, stack.W2t, None, der, EPI_GELU_GRAD)
         dxbuf = _C.batched_gemm_nt(dh_pre, stack.W1t, None, None, EPI_NONE)
-        # weight grads: hand-written TN MFMA kernels write the fp32 grad
-        # buffers in place (accumulate only after the first backward of a step)
         acc = stack.grad_ready
-        _C.batched_gemm_tn(dh_pre, xbuf, stack.main_grad_W1, acc)
-        _C.batched_gemm_tn(dy, h, stack.main_grad_W2, acc)
+        fused = getattr(stack, "_fused_sr_opt", None)
+        if fused is not None and not acc:
+            # fused dW + stochastic-rounding SGD: the TN epilogue applies the
+            # update to W/W^T in place — no fp32 grad buffer round trip
+            # (ffn_gemm_tn.hip; one backward per step by contract)
+            lr, wd = fused.lr, fused.weight_decay
+            seed = fused.sr_seed()
+            _C.fused_dw_sgd_sr(dh_pre, xbuf, stack.W1c, stack.W1t, lr, wd, seed)
+            _C.fused_dw_sgd_sr(dy, h, stack.W2c, stack.W2t, lr, wd, seed ^ 0x55555555)
+        elif fused is not None:
+            raise RuntimeError(
+                "fused dW+SGD supports exactly one backward per optimizer step; "
+                "construct FusedExpertSGD(fused_backward=False) for gradient accumulation"
+            )
+        else:
+            # weight grads: hand-written TN MFMA kernels write the fp32 grad
+            # buffers in place (accumulate only after the first backward of a step)
+            _C.batched_gemm_tn(dh_pre, xbuf, stack.main_grad_W1, acc)
+            _C.batched_gemm_tn(dy, h, stack.main_grad_W2, acc)
         db1 = _C.colsum(dh_pre)
         db2 = _C.colsum(dy)
         if acc:

@@ -10,6 +10,8 @@ void lah_batched_gemm_nt(const void*, const void*, void*, const void*, void*, in
                          int, int, int, hipStream_t);
 void lah_batched_gemm_tn(const void*, const void*, void*, int, int, int, int, int,
                          hipStream_t);
+void lah_batched_gemm_tn_sgd_sr(const void*, const void*, void*, void*, int, int, int,
+                                int, float, float, unsigned, hipStream_t);
 void lah_flash_attn_fwd(const void*, const void*, const void*, void*, void*, void*, int, int,
                         hipStream_t);
 void lah_flash_attn_bwd(const void*, const void*, const void*, const void*, const void*,
@@ -83,6 +85,28 @@ void batched_gemm_tn(torch::Tensor A, torch::Tensor B, torch::Tensor C,
   TORCH_CHECK(I % 128 == 0 && J % 128 == 0, "I/J must be multiples of 128");
   lah_batched_gemm_tn(A.data_ptr(), B.data_ptr(), C.data_ptr(), E, M, I, J,
                       accumulate ? 1 : 0, cur_stream());
+}
+
+// fused dW + stochastic-rounding SGD: W[e] -= lr * (A[e]^T @ B[e] + wd*W[e]),
+// SR-rounded in place, with the transposed bf16 copy refreshed in the same
+// pass — replaces batched_gemm_tn + sgd_step_sr_t for the single-backward-
+// per-step regime (see ffn_gemm_tn.hip)
+void fused_dw_sgd_sr(torch::Tensor A, torch::Tensor B, torch::Tensor W,
+                     torch::Tensor WT, double lr, double weight_decay,
+                     int64_t seed) {
+  CHECK_CUDA(A); CHECK_BF16(A); CHECK_CONTIG(A);
+  CHECK_BF16(B); CHECK_CONTIG(B);
+  CHECK_BF16(W); CHECK_CONTIG(W); CHECK_BF16(WT); CHECK_CONTIG(WT);
+  int E = A.size(0), M = A.size(1), I = A.size(2), J = B.size(2);
+  TORCH_CHECK(B.size(0) == E && B.size(1) == M, "A/B shape mismatch");
+  TORCH_CHECK(W.size(0) == E && W.size(1) == I && W.size(2) == J, "W shape mismatch");
+  TORCH_CHECK(WT.size(0) == E && WT.size(1) == J && WT.size(2) == I, "WT shape mismatch");
+  TORCH_CHECK(M % 64 == 0, "M must be a multiple of 64");
+  TORCH_CHECK(I % 256 == 0 && J % 256 == 0,
+              "fused dW+SGD needs I/J multiples of 256 (use the unfused path)");
+  lah_batched_gemm_tn_sgd_sr(A.data_ptr(), B.data_ptr(), W.data_ptr(), WT.data_ptr(),
+                             E, M, I, J, (float)lr, (float)weight_decay,
+                             (unsigned)seed, cur_stream());
 }
 
 std::vector<torch::Tensor> gating_topk_fwd(torch::Tensor logits, int64_t k) {
@@ -286,6 +310,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("pre") = py::none(), py::arg("epi") = 0);
   m.def("batched_gemm_tn", &batched_gemm_tn, py::arg("A"), py::arg("B"), py::arg("C"),
         py::arg("accumulate") = false);
+  m.def("fused_dw_sgd_sr", &fused_dw_sgd_sr, py::arg("A"), py::arg("B"), py::arg("W"),
+        py::arg("WT"), py::arg("lr"), py::arg("weight_decay") = 0.0,
+        py::arg("seed") = 0);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_fwd_lse", &flash_attn_fwd_lse);
   m.def("flash_attn_bwd", &flash_attn_bwd);

@@ -43,6 +43,23 @@ LAH_DEVINL void glds16(const void* gsrc, void* lds_dst) {
       (__attribute__((address_space(3))) unsigned int*)lds_dst, 16, 0, 0);
 }
 
+
+// --- stochastic-rounding helpers (optim.hip SR-SGD + the fused TN+SGD
+// weight-gradient kernel in ffn_gemm_tn.hip) ---------------------------------
+LAH_DEVINL unsigned hash_u32(unsigned x) {
+  x ^= x >> 16; x *= 0x7feb352du; x ^= x >> 15; x *= 0x846ca68bu; x ^= x >> 16;
+  return x;
+}
+
+LAH_DEVINL __bf16 round_sr(float v, unsigned rnd16) {
+  unsigned bits = __float_as_uint(v);
+  unsigned frac = bits & 0xFFFFu;
+  unsigned trunc = bits & 0xFFFF0000u;
+  if (frac > (rnd16 & 0xFFFFu)) trunc += 0x10000u;
+  unsigned short hi = (unsigned short)(trunc >> 16);
+  return *reinterpret_cast<__bf16*>(&hi);
+}
+
 #define HIP_CHECK(expr)                                                              \
   do {                                                                               \
     hipError_t _e = (expr);                                                          \

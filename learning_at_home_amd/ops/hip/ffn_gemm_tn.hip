@@ -305,3 +305,139 @@ extern "C" void lah_batched_gemm_tn(const void* A, const void* B, void* C, int E
   else
     launch_tn<false>(A, B, C, E, M, I, J, stream);
 }
+
+// ---------------------------------------------------------------------------
+// FUSED dW + stochastic-rounding SGD (round-2).
+//
+// The plain TN kernel is WRITE-bound at bench shapes: it streams a fp32
+// grad buffer (4 B/elem) that the SR-SGD kernel then re-reads (4 B) along
+// with the bf16 weight (2 B) to write W + W^T (4 B) — 14 B/elem of HBM
+// traffic per weight per step. Applying the SR update IN the TN epilogue
+// (legal because the reduction dim M is fully reduced per output tile, so
+// each workgroup owns its dW tile exactly once per backward) cuts that to
+// 6 B/elem: read W, write W and W^T, no grad buffer at all.
+//
+// W^T is produced by staging the updated bf16 tile through LDS (the GEMM's
+// double buffers are dead by then; the 256^2 config's 128 KiB is exactly a
+// 256x256 bf16 tile + padding) and storing coalesced bf16x8 rows of W^T —
+// a direct scatter would hit 64 cache lines per wave-instruction.
+// ---------------------------------------------------------------------------
+template <int WM, int WN, int FM, int FN>
+__launch_bounds__(WM * WN * 64, 1) __global__
+void batched_gemm_tn_sgd_sr_kernel(const __bf16* __restrict__ A,  // [E, M, I]
+                                   const __bf16* __restrict__ B,  // [E, M, J]
+                                   __bf16* __restrict__ W,        // [E, I, J]
+                                   __bf16* __restrict__ WT,       // [E, J, I]
+                                   int M, int I, int J, float lr,
+                                   float weight_decay, unsigned seed) {
+  constexpr int TBI = WM * FM * 16;
+  constexpr int TBJ = WN * FN * 16;
+  constexpr int THREADS = WM * WN * 64;
+  constexpr int TP_LD = TBJ + 8;  // padded row stride for the transpose tile
+  constexpr int GEMM_ELEMS = 2 * (TBI + TBJ) * TBK;
+  constexpr int TRAN_ELEMS = TBI * TP_LD;
+  __shared__ __attribute__((aligned(16)))
+      __bf16 smem[GEMM_ELEMS > TRAN_ELEMS ? GEMM_ELEMS : TRAN_ELEMS];
+  const int HALF = (TBI + TBJ) * TBK;
+
+  const int e = blockIdx.z;
+  const int j0 = blockIdx.x * TBJ;
+  const int i0 = blockIdx.y * TBI;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int wr = wave / WN, wc = wave % WN;
+
+  const __bf16* Ae = A + (long)e * M * I;
+  const __bf16* Be = B + (long)e * M * J;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int a = 0; a < FM; ++a)
+#pragma unroll
+    for (int b = 0; b < FN; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
+
+  const int nk = M / TBK;
+  int cur = 0;
+  stage_tile_t<TBI, THREADS>(Ae, smem, i0, 0, I, tid);
+  stage_tile_t<TBJ, THREADS>(Be, smem + TBI * TBK, j0, 0, J, tid);
+  __syncthreads();
+
+  const int l15 = lane & 15;
+  const int lk8 = (lane >> 4) * 8;
+
+  for (int t = 0; t < nk; ++t) {
+    const __bf16* la = smem + cur * HALF;
+    const __bf16* lb = smem + cur * HALF + TBI * TBK;
+    if (t + 1 < nk) {
+      stage_tile_t<TBI, THREADS>(Ae, smem + (cur ^ 1) * HALF, i0, (t + 1) * TBK, I, tid);
+      stage_tile_t<TBJ, THREADS>(Be, smem + (cur ^ 1) * HALF + TBI * TBK, j0, (t + 1) * TBK, J, tid);
+    }
+#pragma unroll
+    for (int kk = 0; kk < TBK / 32; ++kk) {
+      bf16x8 afrag[FM], bfrag[FN];
+      const int km = kk * 32 + lk8;
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi)
+        afrag[mi] = *(const bf16x8*)(la + TN_OFF(wr * FM * 16 + mi * 16 + l15, km));
+#pragma unroll
+      for (int ni = 0; ni < FN; ++ni)
+        bfrag[ni] = *(const bf16x8*)(lb + TN_OFF(wc * FN * 16 + ni * 16 + l15, km));
+#pragma unroll
+      for (int mi = 0; mi < FM; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < FN; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[mi], bfrag[ni],
+                                                                acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // ---- fused epilogue: SR update of W in place + LDS transpose stage -----
+  __bf16* We = W + (long)e * I * J;
+#pragma unroll
+  for (int mi = 0; mi < FM; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < FN; ++ni) {
+      int lcol = wc * FN * 16 + ni * 16 + l15;
+      int col = j0 + lcol;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int lrow = wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
+        int row = i0 + lrow;
+        long off = (long)row * J + col;
+        float pv = bf2f(We[off]);
+        pv -= lr * (acc[mi][ni][r] + weight_decay * pv);
+        unsigned h = hash_u32((unsigned)(((long)e * I * J + off) & 0xFFFFFFFFu) ^ seed);
+        __bf16 o = round_sr(pv, h >> 8);
+        We[off] = o;
+        smem[lrow * TP_LD + lcol] = o;
+      }
+    }
+  }
+  __syncthreads();
+  // ---- coalesced W^T stores: thread owns 8 consecutive i of one j row ----
+  __bf16* WTe = WT + (long)e * J * I;
+#pragma unroll
+  for (int it = 0; it < (TBI * TBJ) / (THREADS * 8); ++it) {
+    int idx = it * THREADS + tid;       // bf16x8 chunk over [TBJ][TBI/8]
+    int jr = idx / (TBI / 8);           // row of W^T (a column of W)
+    int ic8 = (idx % (TBI / 8)) * 8;    // 8 consecutive W rows
+    bf16x8 v;
+#pragma unroll
+    for (int q = 0; q < 8; ++q) v[q] = smem[(ic8 + q) * TP_LD + jr];
+    *(bf16x8*)(WTe + (long)(j0 + jr) * I + i0 + ic8) = v;
+  }
+}
+
+extern "C" void lah_batched_gemm_tn_sgd_sr(const void* A, const void* B, void* W,
+                                           void* WT, int E, int M, int I, int J,
+                                           float lr, float weight_decay, unsigned seed,
+                                           hipStream_t stream) {
+  // 256^2 tile config only (bench shapes are 256-divisible; bindings check)
+  dim3 grid(J / 256, I / 256, E), block(512);
+  hipLaunchKernelGGL((batched_gemm_tn_sgd_sr_kernel<2, 4, 8, 4>), grid, block, 0,
+                     stream, (const __bf16*)A, (const __bf16*)B, (__bf16*)W,
+                     (__bf16*)WT, M, I, J, lr, weight_decay, seed);
+}

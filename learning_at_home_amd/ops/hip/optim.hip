@@ -87,19 +87,7 @@ __global__ void sgd_step_t_kernel(float* __restrict__ master, const float* __res
 // (E[rounded] = exact value, so small updates are preserved in expectation
 // without an fp32 master copy — 10 B/param of traffic instead of 16) + the
 // transposed bf16 copy in the same pass.
-LAH_DEVINL unsigned hash_u32(unsigned x) {
-  x ^= x >> 16; x *= 0x7feb352du; x ^= x >> 15; x *= 0x846ca68bu; x ^= x >> 16;
-  return x;
-}
-
-LAH_DEVINL __bf16 round_sr(float v, unsigned rnd16) {
-  unsigned bits = __float_as_uint(v);
-  unsigned frac = bits & 0xFFFFu;
-  unsigned trunc = bits & 0xFFFF0000u;
-  if (frac > (rnd16 & 0xFFFFu)) trunc += 0x10000u;
-  unsigned short hi = (unsigned short)(trunc >> 16);
-  return *reinterpret_cast<__bf16*>(&hi);
-}
+// hash_u32 / round_sr moved to common.h (shared with the fused TN+SGD kernel)
 
 __global__ void sgd_step_sr_t_kernel(__bf16* __restrict__ param, const float* __restrict__ grad,
                                      __bf16* __restrict__ paramT, int R, int C, float lr,

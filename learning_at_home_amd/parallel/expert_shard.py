@@ -229,7 +229,8 @@ class FusedExpertSGD:
     refreshed by the tiled transpose kernel. (SURVEY.md §2 hot-op #5.)"""
 
     def __init__(self, stack: ExpertStack, lr: float = 0.01, momentum: float = 0.0,
-                 weight_decay: float = 0.0, stochastic_rounding: bool = False):
+                 weight_decay: float = 0.0, stochastic_rounding: bool = False,
+                 fused_backward: Optional[bool] = None):
         self.stack = stack
         self.lr, self.momentum, self.weight_decay = lr, momentum, weight_decay
         self.stochastic_rounding = stochastic_rounding
@@ -237,6 +238,27 @@ class FusedExpertSGD:
             assert momentum == 0.0, "SR mode supports plain SGD only"
             assert not getattr(stack, "master_weights", True), (
                 "init_gpu_state(master_weights=False) first for SR mode")
+        # fused dW+SGD: the backward's TN epilogue applies the SR update to
+        # W/W^T directly (no fp32 grad buffer; ~14 -> 6 B/elem of HBM per
+        # weight per step). SR mode + 256-aligned dims + one backward/step.
+        if fused_backward is None:
+            import os as _os
+
+            fused_backward = (stochastic_rounding
+                              and stack.d_model % 256 == 0 and stack.ffn_dim % 256 == 0
+                              and _os.environ.get("LAH_FUSED_DW", "1") == "1")
+        self.fused_backward = bool(fused_backward)
+        if self.fused_backward:
+            assert stochastic_rounding, "fused_backward requires stochastic_rounding"
+            assert stack.d_model % 256 == 0 and stack.ffn_dim % 256 == 0, (
+                "fused dW+SGD needs d_model/ffn_dim multiples of 256")
+            stack._fused_sr_opt = self
+            if stack._gpu_state_ready:
+                # the fp32 W grad buffers are never written in fused mode:
+                # release them (flagship: ~8.6 GB of HBM per layer)
+                dev = stack.main_grad_W1.device
+                stack.main_grad_W1 = torch.empty(0, device=dev)
+                stack.main_grad_W2 = torch.empty(0, device=dev)
         self.momentum_bufs = None
         if momentum > 0:
             self.momentum_bufs = [
@@ -245,13 +267,20 @@ class FusedExpertSGD:
             ]
         self.step_count = 0
 
+    def sr_seed(self) -> int:
+        return (self.step_count * 2654435761) & 0x7FFFFFFF
+
     @torch.no_grad()
     def step(self) -> None:
         st = self.stack
         if not st.grad_ready:
             return
-        if self.stochastic_rounding:
-            seed = (self.step_count * 2654435761) & 0x7FFFFFFF
+        if self.fused_backward:
+            # W1/W2 were updated inside the backward's TN epilogue; only the
+            # bias updates (tiny tensors, colsum grads) remain
+            pass
+        elif self.stochastic_rounding:
+            seed = self.sr_seed()
             _C.sgd_step_sr_t(st.W1c, st.main_grad_W1, st.W1t, self.lr,
                              self.weight_decay, seed)
             _C.sgd_step_sr_t(st.W2c, st.main_grad_W2, st.W2t, self.lr,

@@ -382,6 +382,64 @@ class TestMoEOnGPU:
         assert_close_bf16(y_gpu, y_cpu, rtol=5e-2, what="moe gpu vs cpu")
 
 
+class TestFusedDwSgdSr:
+    def test_matches_unfused_within_one_ulp(self):
+        """fused_dw_sgd_sr computes the SAME pre-rounding update as
+        TN-grad + SR-SGD (identical fp32 GEMM accumulation); the stochastic
+        rounding picks one of the two bf16 neighbours, so every element must
+        be within one bf16 ulp of the exact fp32 update."""
+        torch.manual_seed(17)
+        E, M, I, J = 4, 192, 256, 512
+        lr, wd = 0.05, 0.01
+        A = (0.1 * torch.randn(E, M, I, device=DEV)).bfloat16()
+        B = (0.1 * torch.randn(E, M, J, device=DEV)).bfloat16()
+        W = torch.randn(E, I, J, device=DEV).bfloat16().contiguous()
+        WT = W.transpose(1, 2).contiguous()
+        W_fused = W.clone()
+        WT_fused = WT.clone()
+        _C.fused_dw_sgd_sr(A, B, W_fused, WT_fused, lr, wd, seed=123)
+
+        # exact fp32 reference of the pre-rounding value
+        g = torch.einsum("emi,emj->eij", A.float(), B.float())
+        pv = W.float() - lr * (g + wd * W.float())
+        ulp = pv.abs().clamp(min=1e-30) * 2.0**-8 + 1e-12
+        err = (W_fused.float() - pv).abs()
+        assert (err <= ulp * 1.01).all(), (
+            f"fused update not an SR rounding of the exact value: "
+            f"max err/ulp {(err / ulp).max().item():.3f}")
+        # transposed copy is exactly the rounded W, transposed
+        assert torch.equal(WT_fused, W_fused.transpose(1, 2).contiguous())
+        # rounding is not systematically biased to one side (it's stochastic)
+        moved = (W_fused.float() != pv).float().mean().item()
+        assert moved > 0.5, "suspiciously many exact roundings"
+
+    def test_end_to_end_fused_backward_updates_weights(self):
+        """ExpertParallelMoE + FusedExpertSGD(fused_backward): the backward
+        itself applies the W update; step() only touches biases."""
+        torch.manual_seed(18)
+        moe = ExpertParallelMoE(d_model=256, ffn_dim=512, num_experts=8, k=2,
+                                capacity_factor=2.0, aux_loss_coef=0.0).to(DEV)
+        moe.experts.init_gpu_state(master_weights=False)
+        opt = FusedExpertSGD(moe.experts, lr=0.05, stochastic_rounding=True)
+        assert opt.fused_backward, "fused path should be on by default in SR mode"
+        w1_before = moe.experts.W1c.clone()
+        x = torch.randn(4, 64, 256, device=DEV)
+        y = moe(x)
+        y.float().pow(2).mean().backward()
+        torch.cuda.synchronize()
+        assert not torch.equal(moe.experts.W1c, w1_before), (
+            "backward did not apply the fused weight update")
+        assert torch.equal(moe.experts.W1t,
+                           moe.experts.W1c.transpose(1, 2).contiguous())
+        b1_before = moe.experts.W1_bias.detach().clone()
+        opt.step()
+        assert not torch.equal(moe.experts.W1_bias.detach(), b1_before), (
+            "step() did not update biases")
+        # second step end-to-end still healthy
+        y2 = moe(x.detach())
+        assert torch.isfinite(y2.float()).all()
+
+
 def test_flash_attention_fwd_vs_sdpa():
     """Experimental causal flash-attn forward vs fp32 torch sdpa."""
     import torch.nn.functional as F
