@@ -394,40 +394,71 @@ void batched_gemm_tn_sgd_sr_kernel(const __bf16* __restrict__ A,  // [E, M, I]
     cur ^= 1;
   }
 
-  // ---- fused epilogue: SR update of W in place + LDS transpose stage -----
+  // ---- fused epilogue, quartered and VECTOR-shaped --------------------
+  // The C-fragment layout puts consecutive J-columns in different lanes, so
+  // a direct read-modify-write of W would be 2-byte scalar global ops (the
+  // first cut of this kernel did that and LOST 65 ms/step to it). Instead,
+  // per 64-row quarter: (A) scatter acc into an LDS f32 image (cheap DS
+  // scalars), (B) every thread updates 8 CONSECUTIVE W elements — bf16x8
+  // global read + LDS f32 reads + SR + bf16x8 global write — staging the
+  // updated row chunk into a bf16 LDS tile, (C) store W^T rows coalesced
+  // (bf16x8 of 8 consecutive W-rows per j) from that tile.
+  constexpr int QR = 64;                 // quarter rows
+  constexpr int GLD = TBJ + 4;           // f32 image stride (16B-aligned rows)
+  constexpr int OLD_ = TBJ + 8;          // bf16 out-tile stride
+  float* gq = reinterpret_cast<float*>(smem);            // [QR][GLD] f32
+  __bf16* oq = smem + 2 * QR * GLD + 16;                 // [QR][OLD_] bf16
   __bf16* We = W + (long)e * I * J;
+  __bf16* WTe = WT + (long)e * J * I;
+  const int r0l = (lane >> 4) * 4;
+  for (int q4 = 0; q4 < TBI / QR; ++q4) {
+    // (A) fragment scatter: only the fragments whose rows fall in this
+    // quarter participate (barriers stay outside the conditionals)
 #pragma unroll
-  for (int mi = 0; mi < FM; ++mi) {
+    for (int mi = 0; mi < FM; ++mi) {
+      int lrow = wr * FM * 16 + mi * 16;
+      if (lrow < q4 * QR || lrow >= (q4 + 1) * QR) continue;
 #pragma unroll
-    for (int ni = 0; ni < FN; ++ni) {
-      int lcol = wc * FN * 16 + ni * 16 + l15;
-      int col = j0 + lcol;
+      for (int ni = 0; ni < FN; ++ni) {
+        int lcol = wc * FN * 16 + ni * 16 + l15;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int lrow = wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
-        int row = i0 + lrow;
-        long off = (long)row * J + col;
-        float pv = bf2f(We[off]);
-        pv -= lr * (acc[mi][ni][r] + weight_decay * pv);
-        unsigned h = hash_u32((unsigned)(((long)e * I * J + off) & 0xFFFFFFFFu) ^ seed);
-        __bf16 o = round_sr(pv, h >> 8);
-        We[off] = o;
-        smem[lrow * TP_LD + lcol] = o;
+        for (int r = 0; r < 4; ++r)
+          gq[(lrow - q4 * QR + r0l + r) * GLD + lcol] = acc[mi][ni][r];
       }
     }
-  }
-  __syncthreads();
-  // ---- coalesced W^T stores: thread owns 8 consecutive i of one j row ----
-  __bf16* WTe = WT + (long)e * J * I;
+    __syncthreads();
+    // (B) vector update of 8 consecutive W elements per thread-iteration
 #pragma unroll
-  for (int it = 0; it < (TBI * TBJ) / (THREADS * 8); ++it) {
-    int idx = it * THREADS + tid;       // bf16x8 chunk over [TBJ][TBI/8]
-    int jr = idx / (TBI / 8);           // row of W^T (a column of W)
-    int ic8 = (idx % (TBI / 8)) * 8;    // 8 consecutive W rows
-    bf16x8 v;
+    for (int it = 0; it < (QR * TBJ) / (THREADS * 8); ++it) {
+      int idx = it * THREADS + tid;
+      int rr = idx / (TBJ / 8);
+      int c8 = (idx % (TBJ / 8)) * 8;
+      long off = (long)(i0 + q4 * QR + rr) * J + j0 + c8;
+      bf16x8 w8 = *(const bf16x8*)(We + off);
+      bf16x8 o8;
+      unsigned h = hash_u32((unsigned)(((long)e * I * J + off) & 0xFFFFFFFFu) ^ seed);
 #pragma unroll
-    for (int q = 0; q < 8; ++q) v[q] = smem[(ic8 + q) * TP_LD + jr];
-    *(bf16x8*)(WTe + (long)(j0 + jr) * I + i0 + ic8) = v;
+      for (int qq = 0; qq < 8; ++qq) {
+        float pv = bf2f(w8[qq]);
+        pv -= lr * (gq[rr * GLD + c8 + qq] + weight_decay * pv);
+        o8[qq] = round_sr(pv, hash_u32(h + qq) >> 8);
+      }
+      *(bf16x8*)(We + off) = o8;
+      *(bf16x8*)(oq + rr * OLD_ + c8) = o8;
+    }
+    __syncthreads();
+    // (C) coalesced W^T stores: 8 consecutive W-rows of one column
+#pragma unroll
+    for (int it = 0; it < (QR * TBJ) / (THREADS * 8); ++it) {
+      int idx = it * THREADS + tid;
+      int jr = idx / (QR / 8);
+      int ic8 = (idx % (QR / 8)) * 8;
+      bf16x8 v;
+#pragma unroll
+      for (int qq = 0; qq < 8; ++qq) v[qq] = oq[(ic8 + qq) * OLD_ + jr];
+      *(bf16x8*)(WTe + (long)(j0 + jr) * I + i0 + q4 * QR + ic8) = v;
+    }
+    __syncthreads();  // oq/gq reused by the next quarter
   }
 }
 
