@@ -405,7 +405,8 @@ class TestFusedDwSgdSr:
         # exact fp32 reference of the pre-rounding value
         g = torch.einsum("emi,emj->eij", A.float(), B.float())
         pv = W.float() - lr * (g + wd * W.float())
-        ulp = pv.abs().clamp(min=1e-30) * 2.0**-8 + 1e-12
+        # exact bf16 spacing at pv: 2^(floor(log2|pv|) - 8)
+        ulp = torch.pow(2.0, torch.floor(torch.log2(pv.abs().clamp(min=1e-30))) - 8) + 1e-12
         err = (W_fused.float() - pv).abs()
         assert (err <= ulp * 1.01).all(), (
             f"fused update not an SR rounding of the exact value: "
@@ -426,7 +427,7 @@ class TestFusedDwSgdSr:
         opt = FusedExpertSGD(moe.experts, lr=0.05, stochastic_rounding=True)
         assert opt.fused_backward, "fused path should be on by default in SR mode"
         w1_before = moe.experts.W1c.clone()
-        x = torch.randn(4, 64, 256, device=DEV)
+        x = torch.randn(4, 64, 256, device=DEV, requires_grad=True)
         y = moe(x)
         y.float().pow(2).mean().backward()
         torch.cuda.synchronize()
