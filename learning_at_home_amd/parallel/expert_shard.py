@@ -241,12 +241,18 @@ class FusedExpertSGD:
         # fused dW+SGD: the backward's TN epilogue applies the SR update to
         # W/W^T directly (no fp32 grad buffer; ~14 -> 6 B/elem of HBM per
         # weight per step). SR mode + 256-aligned dims + one backward/step.
+        # MEASURED REGIME (profiles/fused_dw_r02.md): wins at small
+        # per-expert batches (M<=~512: 2.82 vs 3.28 ms at M=192) where the
+        # pair is write-bound, LOSES at the flagship's M=2304 (12.5 vs ~9.7
+        # ms: the GEMM turns compute-bound and the in-kernel epilogue is
+        # slower than the streaming SR kernel) -> default OFF; opt in with
+        # LAH_FUSED_DW=1 for small-batch / high-expert-count regimes.
         if fused_backward is None:
             import os as _os
 
             fused_backward = (stochastic_rounding
                               and stack.d_model % 256 == 0 and stack.ffn_dim % 256 == 0
-                              and _os.environ.get("LAH_FUSED_DW", "1") == "1")
+                              and _os.environ.get("LAH_FUSED_DW", "0") == "1")
         self.fused_backward = bool(fused_backward)
         if self.fused_backward:
             assert stochastic_rounding, "fused_backward requires stochastic_rounding"
