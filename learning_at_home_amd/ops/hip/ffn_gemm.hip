@@ -45,15 +45,28 @@ LAH_DEVINL void stage_tile(const __bf16* __restrict__ src, __bf16* lds, int row0
   }
 }
 
-// gelu and its derivative from one tanh evaluation
+// gelu and its derivative from one tanh evaluation. tanh via raw v_exp_f32:
+// t = 1 - 2/(e^{2u}+1). libm tanhf's range/denorm fixup is ~40 cycles per
+// call and dominated the EPI_BIAS_GELU epilogue (PMC: 31.5% ACTIVE_INST on
+// fwd1 vs 16% on the other variants); the exp2 form is 1-2 f32 ulp off
+// libm in the gelu-relevant range, invisible at bf16 output precision.
+// Saturation: u -> +inf gives e=inf -> t=1; u -> -inf gives e=0 -> t=-1.
 LAH_DEVINL void gelu_both(float x, float* g, float* dg) {
   float x2 = x * x;
   float u = GELU_C0 * (x + GELU_C1 * x * x2);
-  float t = tanhf(u);
+  float e = __builtin_amdgcn_exp2f(u * 2.885390081777927f);  // 2*log2(e)
+  float t = 1.0f - 2.0f * __builtin_amdgcn_rcpf(e + 1.0f);
   float du = GELU_C0 * (1.0f + 3.0f * GELU_C1 * x2);
   *g = 0.5f * x * (1.0f + t);
   *dg = 0.5f * (1.0f + t) + 0.5f * x * (1.0f - t * t) * du;
 }
+
+// padded fp32-tile column map: a 4-float gap every 16 columns makes both
+// the b128 read groups (20k+4q mod 64 distinct per lane, CSTR ≡ 0 mod 64)
+// and the b32 scatter writes (<=2-way, free) conflict-free — the unpadded
+// layout was a 4-way read conflict worth 6.6e8 SQ_LDS_BANK_CONFLICT on the
+// fwd1 epilogue (PMC, flagship shapes)
+#define CT_COL(c) ((c) + ((((c) >> 4)) << 2))
 
 // coalesced bf16x8 store pass: ERB rows of the fp32 LDS tile -> global
 template <int EPI, int BN, int CSTR, int THREADS>
@@ -61,6 +74,7 @@ LAH_DEVINL void emit_rows(const float* ctile, __bf16* Ce, __bf16* De, const floa
                           int row0, int n0, int M, int N, int nrows, int tid) {
   constexpr int TPR = BN / 16;  // threads per row (16 cols x bf16x8 pair each)
   const int ecol = (tid % TPR) * 16;
+  const int lcol = CT_COL(ecol);
   // bias is row-invariant: load this thread's 16-wide chunk ONCE
   float bv[16];
   if (EPI == EPI_BIAS || EPI == EPI_BIAS_GELU) {
@@ -72,6 +86,11 @@ LAH_DEVINL void emit_rows(const float* ctile, __bf16* Ce, __bf16* De, const floa
     int row = row0 + rr;
     if (row >= M) continue;
     long goff = (long)row * N + n0 + ecol;
+    // vectorized conflict-free tile reads: 4 x f32x4 per 16-wide chunk
+    float cv[16];
+#pragma unroll
+    for (int q4 = 0; q4 < 4; ++q4)
+      *(f32x4*)(cv + q4 * 4) = *(const f32x4*)(ctile + rr * CSTR + lcol + q4 * 4);
     bf16x8 outv[2], derv[2];
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
@@ -79,7 +98,7 @@ LAH_DEVINL void emit_rows(const float* ctile, __bf16* Ce, __bf16* De, const floa
       if (EPI == EPI_GELU_GRAD) dv = *(const bf16x8*)(De + goff + h * 8);
 #pragma unroll
       for (int q = 0; q < 8; ++q) {
-        float v = ctile[rr * CSTR + ecol + h * 8 + q];
+        float v = cv[h * 8 + q];
         if (EPI == EPI_BIAS) {
           outv[h][q] = f2bf(v + bv[h * 8 + q]);
         } else if (EPI == EPI_BIAS_GELU) {
@@ -208,7 +227,7 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
   // fp32 LDS image with a +4-float row pad: an unpadded BN-stride put every
   // store-pass lane group on 2 banks (16-way; PMC: 100M SQ_LDS_BANK_CONFLICT
   // on the fwd1 epilogue); stride BN+4 spreads rows by one bank.
-  constexpr int CSTR = BN + 4;
+  constexpr int CSTR = BN + (BN >> 4) * 4;  // CT_COL-padded stride
   constexpr int LDS_BYTES = 2 * (BM + BN) * BK * 2;
   constexpr int ERB0 = LDS_BYTES / (CSTR * 4);
   constexpr int ERB = (ERB0 >= BM) ? BM : (ERB0 >= 128 ? 128 : (ERB0 >= 64 ? 64 : 32));
@@ -225,7 +244,7 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int row = wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
-          if (row / ERB == ep) ctile[(row % ERB) * CSTR + col] = acc[mi][ni][r];
+          if (row / ERB == ep) ctile[(row % ERB) * CSTR + CT_COL(col)] = acc[mi][ni][r];
         }
       }
     }
@@ -402,7 +421,7 @@ void batched_gemm_nt_8p_kernel(const __bf16* __restrict__ A,  // [E, M, K]
   }
 
   // epilogue identical to the main kernel (fp32 LDS image, padded)
-  constexpr int CSTR = BN + 4;
+  constexpr int CSTR = BN + (BN >> 4) * 4;  // CT_COL-padded stride
   constexpr int ERB = 64;
   float* ctile = (float*)smem;
 #pragma unroll
@@ -416,7 +435,7 @@ void batched_gemm_nt_8p_kernel(const __bf16* __restrict__ A,  // [E, M, K]
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           int row = wr * FM * 16 + mi * 16 + (lane >> 4) * 4 + r;
-          if (row / ERB == ep) ctile[(row % ERB) * CSTR + col] = acc[mi][ni][r];
+          if (row / ERB == ep) ctile[(row % ERB) * CSTR + CT_COL(col)] = acc[mi][ni][r];
         }
       }
     }
