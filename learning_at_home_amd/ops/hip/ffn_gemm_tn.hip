@@ -29,14 +29,27 @@ template <int COLS, int THREADS>
 LAH_DEVINL void stage_tile_t(const __bf16* __restrict__ src, __bf16* lds, int col0,
                              int m0, int ld, int tid) {
   constexpr int CPR = COLS / 8;  // 16-byte chunks per m-row
+  // 4 m-rows per thread-block so the transposed scatter packs 4 adjacent
+  // m-elements (same i, m4..m4+3 — contiguous under TN_OFF since m&7 stays
+  // within one 8-chunk) into ONE 8-byte ds_write_b64: 8 wide writes per
+  // 4 chunks instead of 32 scalar b16 writes. The scalar transpose scatter
+  // was the TN kernel's documented limiter at large M.
 #pragma unroll
-  for (int it = 0; it < (TBK * COLS) / (THREADS * 8); ++it) {
-    int idx = it * THREADS + tid;  // 16-byte chunk over [TBK][COLS]
-    int m = idx / CPR;
+  for (int it = 0; it < (TBK / 4 * CPR) / THREADS; ++it) {
+    int idx = it * THREADS + tid;
+    int m4 = (idx / CPR) * 4;
     int ic = idx % CPR;
-    bf16x8 v = *(const bf16x8*)(src + (long)(m0 + m) * ld + col0 + ic * 8);
+    bf16x8 v[4];
 #pragma unroll
-    for (int q = 0; q < 8; ++q) lds[TN_OFF(ic * 8 + q, m)] = v[q];
+    for (int r = 0; r < 4; ++r)
+      v[r] = *(const bf16x8*)(src + (long)(m0 + m4 + r) * ld + col0 + ic * 8);
+#pragma unroll
+    for (int q = 0; q < 8; ++q) {
+      bf16x4 pack;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) pack[r] = v[r][q];
+      *(bf16x4*)(lds + TN_OFF(ic * 8 + q, m4)) = pack;
+    }
   }
 }
 
