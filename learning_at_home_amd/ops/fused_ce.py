@@ -17,6 +17,12 @@ from __future__ import annotations
 
 import torch
 
+from . import _C, HAS_HIP
+
+
+def _use_ce_kernels(x: torch.Tensor, weight: torch.Tensor) -> bool:
+    return HAS_HIP and x.is_cuda and weight.shape[0] % 4 == 0
+
 
 class _ChunkedLinearCE(torch.autograd.Function):
     @staticmethod
@@ -25,27 +31,54 @@ class _ChunkedLinearCE(torch.autograd.Function):
         T = x.shape[0]
         total = torch.zeros((), dtype=torch.float32, device=x.device)
         n_valid = int((labels != ignore_index).sum().item())
+        use_hip = _use_ce_kernels(x, weight)
+        labels_i32 = labels.int().contiguous() if use_hip else None
+        lse_all = torch.empty(T, dtype=torch.float32, device=x.device) if use_hip else None
         for s in range(0, T, chunk):
             xs = x[s : s + chunk]
             logits = (xs @ weight.t()).float()
-            total += torch.nn.functional.cross_entropy(
-                logits, labels[s : s + chunk], reduction="sum",
-                ignore_index=ignore_index,
-            )
-        ctx.save_for_backward(x, weight, labels)
+            if use_hip:
+                # one logits read: online lse + label gather (ce.hip); the
+                # torch path materializes log-softmax over [chunk, V] instead
+                lse, picked = _C.ce_rowstats(logits, labels_i32[s : s + chunk],
+                                             ignore_index)
+                lse_all[s : s + chunk] = lse
+                total += (lse - picked).sum()
+            else:
+                total += torch.nn.functional.cross_entropy(
+                    logits, labels[s : s + chunk], reduction="sum",
+                    ignore_index=ignore_index,
+                )
+        ctx.save_for_backward(x, weight, labels,
+                              lse_all if use_hip else torch.empty(0))
         ctx.chunk = chunk
         ctx.ignore_index = ignore_index
         ctx.n_valid = n_valid
+        ctx.use_hip = use_hip
         return total / max(n_valid, 1)
 
     @staticmethod
     def backward(ctx, dloss: torch.Tensor):
-        x, weight, labels = ctx.saved_tensors
+        x, weight, labels, lse_all = ctx.saved_tensors
         T = x.shape[0]
         chunk = ctx.chunk
         ignore_index = ctx.ignore_index
         dx = torch.empty_like(x)
         dw = torch.zeros_like(weight, dtype=torch.float32)
+        if ctx.use_hip:
+            labels_i32 = labels.int().contiguous()
+            # scalar sync once per backward: dloss is a 0-dim tensor (1.0 in
+            # ordinary training); the fused bwd kernel takes gscale by value
+            gs = float(dloss) / max(ctx.n_valid, 1)
+            for s in range(0, T, chunk):
+                xs = x[s : s + chunk]
+                logits = (xs @ weight.t()).float()
+                dlogits = _C.ce_bwd(logits, labels_i32[s : s + chunk],
+                                    lse_all[s : s + chunk].contiguous(), gs,
+                                    ignore_index).to(x.dtype)
+                dx[s : s + chunk] = dlogits @ weight
+                dw += (dlogits.t() @ xs).float()
+            return dx, dw.to(weight.dtype), None, None, None
         scale = dloss.float() / max(ctx.n_valid, 1)
         for s in range(0, T, chunk):
             xs = x[s : s + chunk]

@@ -36,6 +36,9 @@ void lah_sgd_step_t(void*, const void*, void*, void*, int, int, int, float, floa
 void lah_sgd_step_sr_t(void*, const void*, void*, int, int, int, float, float,
                        unsigned, hipStream_t);
 void lah_transpose_bf16(const void*, void*, int, int, int, hipStream_t);
+void lah_ce_rowstats(const void*, const void*, void*, void*, int, long, int, hipStream_t);
+void lah_ce_bwd(const void*, const void*, const void*, void*, float, int, long, int,
+                hipStream_t);
 void lah_colsum(const void*, void*, int, int, int, hipStream_t);
 }
 
@@ -107,6 +110,35 @@ void fused_dw_sgd_sr(torch::Tensor A, torch::Tensor B, torch::Tensor W,
   lah_batched_gemm_tn_sgd_sr(A.data_ptr(), B.data_ptr(), W.data_ptr(), WT.data_ptr(),
                              E, M, I, J, (float)lr, (float)weight_decay,
                              (unsigned)seed, cur_stream());
+}
+
+// chunked lm_head CE row stats: (lse, picked) per row, one logits read
+std::vector<torch::Tensor> ce_rowstats(torch::Tensor logits, torch::Tensor labels,
+                                       int64_t ignore_index) {
+  CHECK_CUDA(logits); CHECK_F32(logits); CHECK_CONTIG(logits);
+  CHECK_I32(labels); CHECK_CONTIG(labels);
+  int T = logits.size(0);
+  long V = logits.size(1);
+  TORCH_CHECK(labels.size(0) == T, "labels/logits mismatch");
+  TORCH_CHECK(V % 4 == 0, "V must be a multiple of 4");
+  auto lse = torch::empty({T}, logits.options());
+  auto picked = torch::empty({T}, logits.options());
+  lah_ce_rowstats(logits.data_ptr(), labels.data_ptr(), lse.data_ptr(),
+                  picked.data_ptr(), T, V, (int)ignore_index, cur_stream());
+  return {lse, picked};
+}
+
+torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
+                     double gscale, int64_t ignore_index) {
+  CHECK_CUDA(logits); CHECK_F32(logits); CHECK_CONTIG(logits);
+  CHECK_I32(labels); CHECK_CONTIG(labels); CHECK_F32(lse); CHECK_CONTIG(lse);
+  int T = logits.size(0);
+  long V = logits.size(1);
+  TORCH_CHECK(V % 4 == 0, "V must be a multiple of 4");
+  auto d = torch::empty({T, V}, logits.options().dtype(at::kBFloat16));
+  lah_ce_bwd(logits.data_ptr(), labels.data_ptr(), lse.data_ptr(), d.data_ptr(),
+             (float)gscale, T, V, (int)ignore_index, cur_stream());
+  return d;
 }
 
 std::vector<torch::Tensor> gating_topk_fwd(torch::Tensor logits, int64_t k) {
@@ -313,6 +345,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_dw_sgd_sr", &fused_dw_sgd_sr, py::arg("A"), py::arg("B"), py::arg("W"),
         py::arg("WT"), py::arg("lr"), py::arg("weight_decay") = 0.0,
         py::arg("seed") = 0);
+  m.def("ce_rowstats", &ce_rowstats, py::arg("logits"), py::arg("labels"),
+        py::arg("ignore_index") = -100);
+  m.def("ce_bwd", &ce_bwd, py::arg("logits"), py::arg("labels"), py::arg("lse"),
+        py::arg("gscale"), py::arg("ignore_index") = -100);
   m.def("flash_attn_fwd", &flash_attn_fwd);
   m.def("flash_attn_fwd_lse", &flash_attn_fwd_lse);
   m.def("flash_attn_bwd", &flash_attn_bwd);

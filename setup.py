@@ -21,6 +21,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "dispatch.hip"),
         os.path.join(HIP_DIR, "optim.hip"),
         os.path.join(HIP_DIR, "attention.hip"),
+        os.path.join(HIP_DIR, "ce.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
