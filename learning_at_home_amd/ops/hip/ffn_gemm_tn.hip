@@ -25,32 +25,46 @@
 //   chunk' = (m>>3) ^ (i & 7) ^ ((i >> 3) & 7)
 #define TN_OFF(i, m) ((i) * TBK + (((((m) >> 3) ^ ((i) & 7) ^ (((i) >> 3) & 7)) << 3) | ((m) & 7)))
 
+// Split staging (guide T14 / G15): reads into registers and the transposed
+// LDS write pass are SEPARATE so the global-read latency of tile t+1 hides
+// under tile t's MFMA section instead of stalling in a vmcnt wait right
+// before it. 4 m-rows per thread so the transposed scatter packs 4 adjacent
+// m-elements (same i, m4..m4+3 — contiguous under TN_OFF since m&7 stays
+// within one 8-chunk) into ONE 8-byte ds_write_b64: 8 wide writes per
+// 4 chunks instead of 32 scalar b16 writes (the scalar scatter was the
+// documented limiter; the fused read+write form left HBM latency exposed).
+template <int COLS, int THREADS>
+LAH_DEVINL void read_tile_t(const __bf16* __restrict__ src, bf16x8 (&v)[4], int col0,
+                            int m0, int ld, int tid) {
+  constexpr int CPR = COLS / 8;
+  static_assert((TBK / 4 * CPR) == THREADS, "one 4mx8i block per thread");
+  int m4 = (tid / CPR) * 4;
+  int ic = tid % CPR;
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    v[r] = *(const bf16x8*)(src + (long)(m0 + m4 + r) * ld + col0 + ic * 8);
+}
+
+template <int COLS, int THREADS>
+LAH_DEVINL void write_tile_t(const bf16x8 (&v)[4], __bf16* lds, int tid) {
+  constexpr int CPR = COLS / 8;
+  int m4 = (tid / CPR) * 4;
+  int ic = tid % CPR;
+#pragma unroll
+  for (int q = 0; q < 8; ++q) {
+    bf16x4 pack;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) pack[r] = v[r][q];
+    *(bf16x4*)(lds + TN_OFF(ic * 8 + q, m4)) = pack;
+  }
+}
+
 template <int COLS, int THREADS>
 LAH_DEVINL void stage_tile_t(const __bf16* __restrict__ src, __bf16* lds, int col0,
                              int m0, int ld, int tid) {
-  constexpr int CPR = COLS / 8;  // 16-byte chunks per m-row
-  // 4 m-rows per thread-block so the transposed scatter packs 4 adjacent
-  // m-elements (same i, m4..m4+3 — contiguous under TN_OFF since m&7 stays
-  // within one 8-chunk) into ONE 8-byte ds_write_b64: 8 wide writes per
-  // 4 chunks instead of 32 scalar b16 writes. The scalar transpose scatter
-  // was the TN kernel's documented limiter at large M.
-#pragma unroll
-  for (int it = 0; it < (TBK / 4 * CPR) / THREADS; ++it) {
-    int idx = it * THREADS + tid;
-    int m4 = (idx / CPR) * 4;
-    int ic = idx % CPR;
-    bf16x8 v[4];
-#pragma unroll
-    for (int r = 0; r < 4; ++r)
-      v[r] = *(const bf16x8*)(src + (long)(m0 + m4 + r) * ld + col0 + ic * 8);
-#pragma unroll
-    for (int q = 0; q < 8; ++q) {
-      bf16x4 pack;
-#pragma unroll
-      for (int r = 0; r < 4; ++r) pack[r] = v[r][q];
-      *(bf16x4*)(lds + TN_OFF(ic * 8 + q, m4)) = pack;
-    }
-  }
+  bf16x8 v[4];
+  read_tile_t<COLS, THREADS>(src, v, col0, m0, ld, tid);
+  write_tile_t<COLS, THREADS>(v, lds, tid);
 }
 
 template <bool ACC, int WM, int WN, int FM, int FN>
@@ -84,8 +98,20 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
 
   const int nk = M / TBK;
   int cur = 0;
-  stage_tile_t<TBI, THREADS>(Ae, smem, i0, 0, I, tid);
-  stage_tile_t<TBJ, THREADS>(Be, smem + TBI * TBK, j0, 0, J, tid);
+  // T14 split schedule, one barrier per K-tile: registers always hold tile
+  // t+1's operands while tile t's MFMAs run, so the HBM read latency of the
+  // staging is covered by a whole MFMA section. Per iteration:
+  //   MFMA(buf[cur])  |  ds_write regs -> buf[cur^1] (safe: nobody reads
+  //   cur^1 this iteration)  |  issue reads of tile t+2  |  barrier.
+  bf16x8 vA[4], vB[4];
+  read_tile_t<TBI, THREADS>(Ae, vA, i0, 0, I, tid);
+  read_tile_t<TBJ, THREADS>(Be, vB, j0, 0, J, tid);
+  write_tile_t<TBI, THREADS>(vA, smem, tid);
+  write_tile_t<TBJ, THREADS>(vB, smem + TBI * TBK, tid);
+  if (nk > 1) {
+    read_tile_t<TBI, THREADS>(Ae, vA, i0, TBK, I, tid);
+    read_tile_t<TBJ, THREADS>(Be, vB, j0, TBK, J, tid);
+  }
   __syncthreads();
 
   const int l15 = lane & 15;
@@ -94,10 +120,6 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
   for (int t = 0; t < nk; ++t) {
     const __bf16* la = smem + cur * HALF;
     const __bf16* lb = smem + cur * HALF + TBI * TBK;
-    if (t + 1 < nk) {
-      stage_tile_t<TBI, THREADS>(Ae, smem + (cur ^ 1) * HALF, i0, (t + 1) * TBK, I, tid);
-      stage_tile_t<TBJ, THREADS>(Be, smem + (cur ^ 1) * HALF + TBI * TBK, j0, (t + 1) * TBK, J, tid);
-    }
 #pragma unroll
     for (int kk = 0; kk < TBK / 32; ++kk) {
       bf16x8 afrag[FM], bfrag[FN];
@@ -118,6 +140,14 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
         for (int ni = 0; ni < FN; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag[mi], bfrag[ni],
                                                                 acc[mi][ni], 0, 0, 0);
+    }
+    if (t + 1 < nk) {
+      write_tile_t<TBI, THREADS>(vA, smem + (cur ^ 1) * HALF, tid);
+      write_tile_t<TBJ, THREADS>(vB, smem + (cur ^ 1) * HALF + TBI * TBK, tid);
+      if (t + 2 < nk) {
+        read_tile_t<TBI, THREADS>(Ae, vA, i0, (t + 2) * TBK, I, tid);
+        read_tile_t<TBJ, THREADS>(Be, vB, j0, (t + 2) * TBK, J, tid);
+      }
     }
     __syncthreads();
     cur ^= 1;
