@@ -260,10 +260,12 @@ void batched_gemm_nt_kernel(const __bf16* __restrict__ A,  // [E, M, K]
 
 
 static int lah_xcd_swz() {
+  // default ON since round 2: +1.9% on the flagship step (in-step L2
+  // locality; the isolated NT shapes move less). LAH_XCD_SWZ=0 disables.
   static int v = -1;
   if (v < 0) {
     const char* env = getenv("LAH_XCD_SWZ");
-    v = (env != nullptr && env[0] == '1') ? 1 : 0;
+    v = (env != nullptr && env[0] == '0') ? 0 : 1;
   }
   return v;
 }
