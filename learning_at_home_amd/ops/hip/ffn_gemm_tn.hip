@@ -79,9 +79,17 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
   __shared__ __attribute__((aligned(16))) __bf16 smem[2 * (TBI + TBJ) * TBK];
   const int HALF = (TBI + TBJ) * TBK;
 
-  const int e = blockIdx.z;
-  const int j0 = blockIdx.x * TBJ;
-  const int i0 = blockIdx.y * TBI;
+  // XCD remap (same 64-block window as the NT kernels: +L2 locality for
+  // the A/B row-bands shared by neighbouring tiles)
+  int lin = (blockIdx.z * gridDim.y + blockIdx.y) * gridDim.x + blockIdx.x;
+  {
+    int total = gridDim.x * gridDim.y * gridDim.z;
+    if (lin < (total & ~63))
+      lin = (lin & ~63) | ((lin & 7) << 3) | ((lin >> 3) & 7);
+  }
+  const int e = lin / (gridDim.x * gridDim.y);
+  const int j0 = (lin % gridDim.x) * TBJ;
+  const int i0 = ((lin / gridDim.x) % gridDim.y) * TBI;
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
