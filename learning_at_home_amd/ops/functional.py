@@ -239,17 +239,13 @@ class _BatchedFFNHip(torch.autograd.Function):
                 "fused dW+SGD supports exactly one backward per optimizer step; "
                 "construct FusedExpertSGD(fused_backward=False) for gradient accumulation"
             )
-            db1 = _C.colsum(dh_pre)
-            db2 = _C.colsum(dy)
         else:
             # weight grads: hand-written TN MFMA kernels write the fp32 grad
-            # buffers in place (accumulate only after the first backward of a
-            # step); the bias-grad colsums ride the same kernels' fragment
-            # stream instead of re-reading dh_pre/dy
-            db1 = _C.batched_gemm_tn(dh_pre, xbuf, stack.main_grad_W1, acc,
-                                     with_bias_grad=True)
-            db2 = _C.batched_gemm_tn(dy, h, stack.main_grad_W2, acc,
-                                     with_bias_grad=True)
+            # buffers in place (accumulate only after the first backward of a step)
+            _C.batched_gemm_tn(dh_pre, xbuf, stack.main_grad_W1, acc)
+            _C.batched_gemm_tn(dy, h, stack.main_grad_W2, acc)
+        db1 = _C.colsum(dh_pre)
+        db2 = _C.colsum(dy)
         if acc:
             stack.main_grad_b1.add_(db1)
             stack.main_grad_b2.add_(db2)

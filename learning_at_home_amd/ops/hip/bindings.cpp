@@ -8,8 +8,8 @@
 extern "C" {
 void lah_batched_gemm_nt(const void*, const void*, void*, const void*, void*, int, int,
                          int, int, int, hipStream_t);
-void lah_batched_gemm_tn(const void*, const void*, void*, void*, int, int, int, int,
-                         int, hipStream_t);
+void lah_batched_gemm_tn(const void*, const void*, void*, int, int, int, int, int,
+                         hipStream_t);
 void lah_batched_gemm_tn_sgd_sr(const void*, const void*, void*, void*, int, int, int,
                                 int, float, float, unsigned, hipStream_t);
 void lah_flash_attn_fwd(const void*, const void*, const void*, void*, void*, void*, int, int,
@@ -76,13 +76,9 @@ torch::Tensor batched_gemm_nt(torch::Tensor A, torch::Tensor B,
   return C;
 }
 
-// C[e,i,j] (fp32, written in place) = sum_m A[e,m,i] * B[e,m,j].
-// With with_bias_grad, also returns db[e,i] = sum_m A[e,m,i] computed from
-// the fragments the GEMM already streams (saves the colsum kernel's full
-// re-read of A — ~6 GB/layer at the flagship shapes).
-c10::optional<torch::Tensor> batched_gemm_tn(torch::Tensor A, torch::Tensor B,
-                                             torch::Tensor C, bool accumulate,
-                                             bool with_bias_grad) {
+// C[e,i,j] (fp32, written in place) = sum_m A[e,m,i] * B[e,m,j]
+void batched_gemm_tn(torch::Tensor A, torch::Tensor B, torch::Tensor C,
+                     bool accumulate) {
   CHECK_CUDA(A); CHECK_BF16(A); CHECK_CONTIG(A);
   CHECK_BF16(B); CHECK_CONTIG(B); CHECK_F32(C); CHECK_CONTIG(C);
   int E = A.size(0), M = A.size(1), I = A.size(2), J = B.size(2);
@@ -90,15 +86,8 @@ c10::optional<torch::Tensor> batched_gemm_tn(torch::Tensor A, torch::Tensor B,
   TORCH_CHECK(C.size(0) == E && C.size(1) == I && C.size(2) == J, "C shape mismatch");
   TORCH_CHECK(M % 64 == 0, "M must be a multiple of 64");
   TORCH_CHECK(I % 128 == 0 && J % 128 == 0, "I/J must be multiples of 128");
-  c10::optional<torch::Tensor> db;
-  void* db_p = nullptr;
-  if (with_bias_grad) {
-    db = torch::empty({E, I}, C.options());
-    db_p = db->data_ptr();
-  }
-  lah_batched_gemm_tn(A.data_ptr(), B.data_ptr(), C.data_ptr(), db_p, E, M, I, J,
+  lah_batched_gemm_tn(A.data_ptr(), B.data_ptr(), C.data_ptr(), E, M, I, J,
                       accumulate ? 1 : 0, cur_stream());
-  return db;
 }
 
 // fused dW + stochastic-rounding SGD: W[e] -= lr * (A[e]^T @ B[e] + wd*W[e]),
@@ -352,7 +341,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("A"), py::arg("B"), py::arg("bias") = py::none(),
         py::arg("pre") = py::none(), py::arg("epi") = 0);
   m.def("batched_gemm_tn", &batched_gemm_tn, py::arg("A"), py::arg("B"), py::arg("C"),
-        py::arg("accumulate") = false, py::arg("with_bias_grad") = false);
+        py::arg("accumulate") = false);
   m.def("fused_dw_sgd_sr", &fused_dw_sgd_sr, py::arg("A"), py::arg("B"), py::arg("W"),
         py::arg("WT"), py::arg("lr"), py::arg("weight_decay") = 0.0,
         py::arg("seed") = 0);

@@ -72,7 +72,6 @@ __launch_bounds__(WM * WN * 64, (WM * WN > 4) ? 1 : 2) __global__
 void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
                             const __bf16* __restrict__ B,  // [E, M, J]
                             float* __restrict__ C,         // [E, I, J]
-                            float* __restrict__ DBIAS,     // [E, I] or null
                             int M, int I, int J) {
   constexpr int TBI = WM * FM * 16;
   constexpr int TBJ = WN * FN * 16;
@@ -104,14 +103,6 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
   for (int a = 0; a < FM; ++a)
 #pragma unroll
     for (int b = 0; b < FN; ++b) acc[a][b] = {0.f, 0.f, 0.f, 0.f};
-
-  // fused bias-grad colsum (db[i] = sum_m A[m][i]): the j0==0 / wc==0
-  // blocks already stream every A element through their fragments — summing
-  // them here saves the separate colsum kernel's full re-read of A
-  const bool do_db = (DBIAS != nullptr) && (j0 == 0) && (wc == 0);
-  float asum[FM];
-#pragma unroll
-  for (int i = 0; i < FM; ++i) asum[i] = 0.f;
 
   const int nk = M / TBK;
   int cur = 0;
@@ -151,12 +142,6 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
         int row = wc * FN * 16 + ni * 16 + l15;
         bfrag[ni] = *(const bf16x8*)(lb + TN_OFF(row, km));
       }
-      if (do_db) {
-#pragma unroll
-        for (int mi = 0; mi < FM; ++mi)
-#pragma unroll
-          for (int j = 0; j < 8; ++j) asum[mi] += bf2f(afrag[mi][j]);
-      }
 #pragma unroll
       for (int mi = 0; mi < FM; ++mi)
 #pragma unroll
@@ -174,19 +159,6 @@ void batched_gemm_tn_kernel(const __bf16* __restrict__ A,  // [E, M, I]
     }
     __syncthreads();
     cur ^= 1;
-  }
-
-  if (do_db) {
-    // lane (l15, h) holds partial sums over its m-chunks; fold the 4
-    // h-groups and let the h==0 lanes write row i (one writer per i)
-#pragma unroll
-    for (int mi = 0; mi < FM; ++mi) {
-      float v = asum[mi];
-      v += __shfl_xor(v, 16, 64);
-      v += __shfl_xor(v, 32, 64);
-      if ((lane >> 4) == 0)
-        DBIAS[(long)e * I + i0 + wr * FM * 16 + mi * 16 + l15] = v;
-    }
   }
 
   // direct fragment-mapped epilogue (an LDS-staged variant was A/B-tested
@@ -358,9 +330,9 @@ static int lah_use_tn_tr16() {
 }
 
 template <bool ACC>
-static void launch_tn(const void* A, const void* B, void* C, void* DB, int E, int M,
-                      int I, int J, hipStream_t stream) {
-  if (lah_use_tn_tr16() && DB == nullptr && I % 256 == 0 && J % 256 == 0) {
+static void launch_tn(const void* A, const void* B, void* C, int E, int M, int I, int J,
+                      hipStream_t stream) {
+  if (lah_use_tn_tr16() && I % 256 == 0 && J % 256 == 0) {
     dim3 grid(J / 256, I / 256, E), block(512);
     hipLaunchKernelGGL((batched_gemm_tn_tr_kernel<ACC, 2, 4, 8, 4>), grid, block, 0,
                        stream, (const __bf16*)A, (const __bf16*)B, (float*)C, M, I, J);
@@ -369,21 +341,20 @@ static void launch_tn(const void* A, const void* B, void* C, void* DB, int E, in
   if (I % 256 == 0 && J % 256 == 0) {
     dim3 grid(J / 256, I / 256, E), block(512);
     hipLaunchKernelGGL((batched_gemm_tn_kernel<ACC, 2, 4, 8, 4>), grid, block, 0, stream,
-                       (const __bf16*)A, (const __bf16*)B, (float*)C, (float*)DB, M, I, J);
+                       (const __bf16*)A, (const __bf16*)B, (float*)C, M, I, J);
   } else {
     dim3 grid(J / 128, I / 128, E), block(256);
     hipLaunchKernelGGL((batched_gemm_tn_kernel<ACC, 2, 2, 4, 4>), grid, block, 0, stream,
-                       (const __bf16*)A, (const __bf16*)B, (float*)C, (float*)DB, M, I, J);
+                       (const __bf16*)A, (const __bf16*)B, (float*)C, M, I, J);
   }
 }
 
-extern "C" void lah_batched_gemm_tn(const void* A, const void* B, void* C, void* DB,
-                                    int E, int M, int I, int J, int accumulate,
-                                    hipStream_t stream) {
+extern "C" void lah_batched_gemm_tn(const void* A, const void* B, void* C, int E, int M,
+                                    int I, int J, int accumulate, hipStream_t stream) {
   if (accumulate)
-    launch_tn<true>(A, B, C, DB, E, M, I, J, stream);
+    launch_tn<true>(A, B, C, E, M, I, J, stream);
   else
-    launch_tn<false>(A, B, C, DB, E, M, I, J, stream);
+    launch_tn<false>(A, B, C, E, M, I, J, stream);
 }
 
 // ---------------------------------------------------------------------------
