@@ -59,6 +59,36 @@ def main():
     torch.manual_seed(0)
     model = TransformerDMoE(cfg, moe_factory).eval()
     prompt = torch.randint(0, cfg.vocab_size, (1, args.prompt))
+
+    if args.mode == "graph":
+        # hipGraph-captured decode vs the same static step replayed eagerly
+        # (GPU; per-token latency A/B — graphs remove the launch gaps)
+        from learning_at_home_amd.models import GraphDecoder
+
+        assert torch.cuda.is_available(), "--mode graph needs a GPU"
+        dev = torch.device("cuda:0")
+        model = model.to(dev)
+        for block in model.blocks:
+            if hasattr(block.moe, "experts"):
+                block.moe.experts.float()
+                block.moe.experts.init_gpu_state(master_weights=True)
+        prompt = prompt.to(dev)
+        for use_graph in (False, True):
+            dec = GraphDecoder(model, batch_size=1, use_graph=use_graph)
+            dec.generate(prompt, max_new_tokens=4)  # warmup incl. capture
+            dec2 = GraphDecoder(model, batch_size=1, use_graph=use_graph)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            out = dec2.generate(prompt, max_new_tokens=args.tokens)
+            torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+            n = out.shape[1] - args.prompt
+            tag = "hipGraph replay" if use_graph else "eager static"
+            print(f"mode=graph [{tag}]: {n} tokens in {dt:.3f}s = "
+                  f"{1e3 * dt / n:.2f} ms/token (d={args.d_model}, L={args.layers}, "
+                  f"E={args.experts}, k=2, GPU)")
+        return
+
     model.generate(prompt, max_new_tokens=2)  # warmup
     t0 = time.perf_counter()
     out = model.generate(prompt, max_new_tokens=args.tokens)

@@ -4,6 +4,7 @@ bench.py (BASELINE.json configs #2/#3)."""
 
 from .experts import FFNExpert, TransformerExpert, NopExpert, name_to_block
 from .transformer_dmoe import TransformerDMoE, TransformerDMoEConfig, DMoETransformerBlock
+from .graph_decode import GraphDecoder
 
 __all__ = [
     "FFNExpert",
@@ -13,4 +14,5 @@ __all__ = [
     "TransformerDMoE",
     "TransformerDMoEConfig",
     "DMoETransformerBlock",
+    "GraphDecoder",
 ]

@@ -205,3 +205,23 @@ def test_synthetic_markov_structure_is_learned():
     floor = math.log(V)
     assert first > floor - 0.7  # starts near the unigram floor
     assert last < floor - 0.8, f"no structure learned: {first:.3f} -> {last:.3f} (ln V={floor:.3f})"
+
+
+def test_graph_decoder_static_path_matches_generate_cpu():
+    """GraphDecoder's static-state step (eager on CPU) must reproduce the
+    dynamic-cache model.generate greedy tokens."""
+    import torch as _t
+
+    from learning_at_home_amd.models import GraphDecoder, TransformerDMoE, TransformerDMoEConfig
+
+    _t.manual_seed(5)
+    cfg = TransformerDMoEConfig(vocab_size=97, d_model=32, n_heads=2, n_layers=2,
+                                seq_len=40, num_experts=1, k_best=1, ffn_dim=64)
+    model = TransformerDMoE(cfg).eval()  # dense FFN fallback
+    prompt = _t.randint(0, 97, (1, 6))
+    with _t.no_grad():
+        ref = model.generate(prompt, max_new_tokens=12)
+    dec = GraphDecoder(model, batch_size=1, use_graph=False)
+    out = dec.generate(prompt, max_new_tokens=12)
+    assert out.shape == ref.shape
+    assert _t.equal(out, ref), (out.tolist(), ref.tolist())

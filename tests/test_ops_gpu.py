@@ -599,3 +599,34 @@ class TestFusedCEKernels:
         assert torch.allclose(x.grad, x2.grad, rtol=5e-2, atol=1e-3), (
             (x.grad - x2.grad).abs().max().item())
         assert torch.allclose(w.grad, w2.grad, rtol=5e-2, atol=1e-3)
+
+
+def test_graph_decoder_replay_matches_eager_static():
+    """hipGraph-replayed decode must produce exactly the eager static-state
+    step's tokens, through the full routed MoE path (gating/dispatch/HIP
+    FFN/combine inside the captured graph)."""
+    from learning_at_home_amd.models import GraphDecoder, TransformerDMoE, TransformerDMoEConfig
+    from learning_at_home_amd.parallel import ExpertParallelMoE
+
+    torch.manual_seed(6)
+    cfg = TransformerDMoEConfig(vocab_size=512, d_model=256, n_heads=4, n_layers=2,
+                                seq_len=64, num_experts=8, k_best=2, ffn_dim=512)
+
+    def moe_factory(i, c):
+        return ExpertParallelMoE(d_model=c.d_model, ffn_dim=c.ffn_dim,
+                                 num_experts=c.num_experts, k=c.k_best,
+                                 router_noise_std=0.0)
+
+    model = TransformerDMoE(cfg, moe_factory).to(DEV).eval()
+    for block in model.blocks:
+        block.moe.experts.float()
+        block.moe.experts.init_gpu_state(master_weights=True)
+    prompt = torch.randint(0, 512, (1, 8), device=DEV)
+
+    dec_eager = GraphDecoder(model, batch_size=1, use_graph=False)
+    out_eager = dec_eager.generate(prompt, max_new_tokens=24)
+    dec_graph = GraphDecoder(model, batch_size=1, use_graph=True)
+    out_graph = dec_graph.generate(prompt, max_new_tokens=24)
+    assert out_graph.shape == out_eager.shape
+    assert torch.equal(out_graph, out_eager), (
+        out_graph.tolist(), out_eager.tolist())
