@@ -21,7 +21,7 @@ from learning_at_home_amd.models import TransformerDMoE, TransformerDMoEConfig
 
 def main():
     p = argparse.ArgumentParser()
-    p.add_argument("--mode", choices=["local", "remote"], default="local")
+    p.add_argument("--mode", choices=["local", "remote", "graph"], default="local")
     p.add_argument("--d-model", type=int, default=256)
     p.add_argument("--layers", type=int, default=2)
     p.add_argument("--experts", type=int, default=4)
