@@ -630,3 +630,47 @@ def test_graph_decoder_replay_matches_eager_static():
     assert out_graph.shape == out_eager.shape
     assert torch.equal(out_graph, out_eager), (
         out_graph.tolist(), out_eager.tolist())
+
+
+def test_rccl_a2a_on_side_stream_world1():
+    """The chunked-a2a pipeline issues RCCL collectives from a side HIP
+    stream with event ordering, and relies on autograd replaying the
+    backward collective on the same stream. Exercise exactly that pattern
+    on a real RCCL communicator (world 1 — the collective degenerates to a
+    device copy, but the stream/communicator mechanics are the ones the
+    driver's 8-GPU run will hit first)."""
+    import os as _os
+    import socket
+
+    import torch.distributed as dist
+
+    from learning_at_home_amd.parallel.expert_shard import _AllToAll
+
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        port = s.getsockname()[1]
+    dist.init_process_group("nccl", rank=0, world_size=1,
+                            init_method=f"tcp://127.0.0.1:{port}")
+    try:
+        dev = torch.device(DEV)
+        x = torch.randn(8, 16, device=dev, requires_grad=True)
+        comm = torch.cuda.Stream(device=dev)
+        ev_in = torch.cuda.Event()
+        ev_in.record()
+        with torch.cuda.stream(comm):
+            comm.wait_event(ev_in)
+            y = _AllToAll.apply(x, None)
+            ev_out = torch.cuda.Event()
+            ev_out.record()
+        cur = torch.cuda.current_stream(dev)
+        cur.wait_event(ev_out)
+        y.record_stream(cur)
+        z = (y.float() ** 2).sum()
+        z.backward()  # backward a2a replays on the comm stream (torch
+        torch.cuda.synchronize()
+        assert torch.allclose(x.grad, 2 * x.detach().float().to(x.grad.dtype), atol=1e-5)
+        assert torch.allclose(y.detach(), x.detach())
+    finally:
+        dist.destroy_process_group()
