@@ -178,10 +178,15 @@ def main():
     for name, param in model.named_parameters():
         if ".moe.experts." not in name:
             trunk_params.append(param)
+    opt_stream = torch.cuda.Stream(device=device) if use_cuda else None
     for block in model.blocks:
         if device.type == "cuda" and HAS_HIP:
-            expert_opts.append(FusedExpertSGD(block.moe.experts, lr=args.lr,
-                                              stochastic_rounding=not args.master_weights))
+            opt = FusedExpertSGD(block.moe.experts, lr=args.lr,
+                                 stochastic_rounding=not args.master_weights)
+            # overlap the expert update with the NEXT step's forward: layer
+            # L's weights aren't read again until layer L's MoE runs
+            opt.set_side_stream(opt_stream)
+            expert_opts.append(opt)
         else:
             expert_opts.append(torch.optim.SGD(block.moe.experts.parameters(), lr=args.lr))
     # trunk in bf16 needs fp32 masters: plain SGD on bf16 loses sub-ulp updates

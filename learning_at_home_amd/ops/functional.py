@@ -202,9 +202,22 @@ def combine(ybuf, topk_idx, pos, w, inv):
 # ---------------------------------------------------------------------------
 # batched expert FFN (GPU fused path; weight grads -> stack.main_grad_*)
 # ---------------------------------------------------------------------------
+def _wait_pending_opt(stack) -> None:
+    # side-stream optimizer overlap (FusedExpertSGD.set_side_stream): the
+    # previous step's weight update may still be in flight on the side
+    # stream — fence before reading W (forward) or writing grads (backward).
+    # Resolve through slice views to the owning stack so the event clears.
+    owner = getattr(stack, "_parent", stack)
+    ev = getattr(owner, "_opt_done_event", None)
+    if ev is not None:
+        torch.cuda.current_stream().wait_event(ev)
+        owner._opt_done_event = None
+
+
 class _BatchedFFNHip(torch.autograd.Function):
     @staticmethod
     def forward(ctx, xbuf: torch.Tensor, stack):
+        _wait_pending_opt(stack)
         E, C, D = xbuf.shape
         # der = gelu'(pre-act), computed for free in the fwd epilogue (one tanh)
         der = torch.empty(E, C, stack.ffn_dim, dtype=torch.bfloat16, device=xbuf.device)
@@ -218,6 +231,7 @@ class _BatchedFFNHip(torch.autograd.Function):
     def backward(ctx, dy: torch.Tensor):
         xbuf, der, h = ctx.saved_tensors
         stack = ctx.stack
+        _wait_pending_opt(stack)
         dy = dy.contiguous()
         # dh_pre = (dy @ W2t^T) * der; dx = dh_pre @ W1t^T   (MFMA NT)
         # (computed BEFORE any fused weight update below: dx must see the

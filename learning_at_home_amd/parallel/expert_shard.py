@@ -276,8 +276,29 @@ class FusedExpertSGD:
     def sr_seed(self) -> int:
         return (self.step_count * 2654435761) & 0x7FFFFFFF
 
+    def set_side_stream(self, stream: "torch.cuda.Stream") -> None:
+        """Overlap the optimizer pass with the NEXT step's forward: step()
+        launches its kernels on `stream` and records a completion event on
+        the stack; batched_ffn's forward/backward wait that event before
+        touching the weights/grads (layer L's update hides under the next
+        step's pre-L compute — ~23 ms of SR traffic at the flagship)."""
+        self._side_stream = stream
+
     @torch.no_grad()
     def step(self) -> None:
+        side = getattr(self, "_side_stream", None)
+        if side is not None and self.stack.grad_ready:
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                self._step_body()
+            ev = torch.cuda.Event()
+            ev.record(side)
+            self.stack._opt_done_event = ev
+            return
+        self._step_body()
+
+    @torch.no_grad()
+    def _step_body(self) -> None:
         st = self.stack
         if not st.grad_ready:
             return

@@ -674,3 +674,49 @@ def test_rccl_a2a_on_side_stream_world1():
         assert torch.allclose(y.detach(), x.detach())
     finally:
         dist.destroy_process_group()
+
+
+def test_expert_opt_side_stream_overlap_bitwise_equal():
+    """FusedExpertSGD on a side stream (overlapped with the next forward)
+    must produce the SAME training trajectory as the default-stream path —
+    only kernel ordering changes, every data dependency is event-fenced."""
+    from learning_at_home_amd.models import TransformerDMoE, TransformerDMoEConfig
+    from learning_at_home_amd.parallel import ExpertParallelMoE
+
+    def run(overlap):
+        torch.manual_seed(33)
+        cfg = TransformerDMoEConfig(vocab_size=512, d_model=256, n_heads=4,
+                                    n_layers=2, seq_len=64, num_experts=8,
+                                    k_best=2, ffn_dim=512)
+
+        def moe_factory(i, c):
+            return ExpertParallelMoE(d_model=c.d_model, ffn_dim=c.ffn_dim,
+                                     num_experts=c.num_experts, k=c.k_best,
+                                     router_noise_std=0.0)
+
+        model = TransformerDMoE(cfg, moe_factory).to(DEV).to(torch.bfloat16)
+        opts = []
+        stream = torch.cuda.Stream(device=DEV) if overlap else None
+        for block in model.blocks:
+            block.moe.experts.float()
+            block.moe.experts.init_gpu_state(master_weights=False)
+            opt = FusedExpertSGD(block.moe.experts, lr=0.05, stochastic_rounding=True)
+            if overlap:
+                opt.set_side_stream(stream)
+            opts.append(opt)
+        g = torch.Generator(device="cpu").manual_seed(9)
+        losses = []
+        for _ in range(4):
+            ids = torch.randint(0, 512, (2, 65), generator=g).to(DEV)
+            loss = model.loss(ids[:, :-1], ids[:, 1:])
+            loss.backward()
+            for o in opts:
+                o.step()
+            losses.append(loss.detach())
+            model.zero_grad(set_to_none=True)
+        torch.cuda.synchronize()
+        return [float(l) for l in losses]
+
+    base = run(False)
+    over = run(True)
+    assert base == over, (base, over)
