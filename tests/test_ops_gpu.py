@@ -676,47 +676,36 @@ def test_rccl_a2a_on_side_stream_world1():
         dist.destroy_process_group()
 
 
-def test_expert_opt_side_stream_overlap_bitwise_equal():
-    """FusedExpertSGD on a side stream (overlapped with the next forward)
-    must produce the SAME training trajectory as the default-stream path —
-    only kernel ordering changes, every data dependency is event-fenced."""
-    from learning_at_home_amd.models import TransformerDMoE, TransformerDMoEConfig
-    from learning_at_home_amd.parallel import ExpertParallelMoE
-
+def test_expert_opt_side_stream_fence():
+    """FusedExpertSGD on a side stream: (a) the update is bitwise the same
+    as the default-stream path (same kernels, same manually-set grads, same
+    SR seed), and (b) a forward issued right after step() sees the UPDATED
+    weights — the per-stack event fence in batched_ffn orders the streams.
+    (A full-trajectory bitwise comparison is NOT well-defined on this stack:
+    assign_slots uses atomics, so capacity-dropped token sets vary run to
+    run independent of the overlap.)"""
     def run(overlap):
-        torch.manual_seed(33)
-        cfg = TransformerDMoEConfig(vocab_size=512, d_model=256, n_heads=4,
-                                    n_layers=2, seq_len=64, num_experts=8,
-                                    k_best=2, ffn_dim=512)
-
-        def moe_factory(i, c):
-            return ExpertParallelMoE(d_model=c.d_model, ffn_dim=c.ffn_dim,
-                                     num_experts=c.num_experts, k=c.k_best,
-                                     router_noise_std=0.0)
-
-        model = TransformerDMoE(cfg, moe_factory).to(DEV).to(torch.bfloat16)
-        opts = []
-        stream = torch.cuda.Stream(device=DEV) if overlap else None
-        for block in model.blocks:
-            block.moe.experts.float()
-            block.moe.experts.init_gpu_state(master_weights=False)
-            opt = FusedExpertSGD(block.moe.experts, lr=0.05, stochastic_rounding=True)
-            if overlap:
-                opt.set_side_stream(stream)
-            opts.append(opt)
-        g = torch.Generator(device="cpu").manual_seed(9)
-        losses = []
-        for _ in range(4):
-            ids = torch.randint(0, 512, (2, 65), generator=g).to(DEV)
-            loss = model.loss(ids[:, :-1], ids[:, 1:])
-            loss.backward()
-            for o in opts:
-                o.step()
-            losses.append(loss.detach())
-            model.zero_grad(set_to_none=True)
+        torch.manual_seed(44)
+        st = ExpertStack(4, 256, 512).to(DEV)
+        st.init_gpu_state(master_weights=False)
+        opt = FusedExpertSGD(st, lr=0.1, stochastic_rounding=True)
+        if overlap:
+            opt.set_side_stream(torch.cuda.Stream(device=DEV))
+        torch.manual_seed(45)
+        st.main_grad_W1.normal_()
+        st.main_grad_W2.normal_()
+        st.main_grad_b1.normal_()
+        st.main_grad_b2.normal_()
+        st.grad_ready = True
+        opt.step()
+        # forward immediately after: must observe the post-update weights
+        xbuf = torch.ones(4, 64, 256, device=DEV, dtype=torch.bfloat16)
+        y = Fops.batched_ffn(xbuf, st)
         torch.cuda.synchronize()
-        return [float(l) for l in losses]
+        return st.W1c.clone(), st.W1t.clone(), y.clone()
 
-    base = run(False)
-    over = run(True)
-    assert base == over, (base, over)
+    w_base, wt_base, y_base = run(False)
+    w_over, wt_over, y_over = run(True)
+    assert torch.equal(w_base, w_over), "side-stream update differs from default-stream"
+    assert torch.equal(wt_base, wt_over)
+    assert torch.equal(y_base, y_over), "forward after overlapped step saw stale weights"
