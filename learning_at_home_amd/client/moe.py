@@ -68,6 +68,10 @@ class RemoteMixtureOfExperts(nn.Module):
         self._active_cache: Dict[str, Optional[Endpoint]] = {}
         self._active_cache_time = -float("inf")
         self.aux_loss: Optional[torch.Tensor] = None
+        # observability: cumulative backward-RPC quorum
+        # accounting — experts that returned grads vs dropped, and how many
+        # backwards missed backward_k_min entirely (grads silently partial)
+        self.backward_stats = {"experts_ok": 0, "experts_dropped": 0, "quorum_missed": 0}
 
     # ------------------------------------------------------------------
     def _candidate_uids(self, grid_logits: List[torch.Tensor]) -> Tuple[List[List[str]], torch.Tensor]:
@@ -154,7 +158,7 @@ class RemoteMixtureOfExperts(nn.Module):
         y = _RemoteMoECallBatched.apply(
             flat, sel_logits, sel_uids, endpoint_of, self.k_min,
             self.forward_timeout, self.backward_timeout, self.backward_k_min,
-            self.timeout_after_k_min,
+            self.timeout_after_k_min, self.backward_stats,
         )
 
         if self.aux_loss_coef > 0:
@@ -225,7 +229,8 @@ class _RemoteMoECallBatched(torch.autograd.Function):
     def forward(ctx, x: torch.Tensor, sel_logits: torch.Tensor, sel_uids: List[List[str]],
                 endpoint_of: Dict[str, Endpoint], k_min: int,
                 forward_timeout: float, backward_timeout: float,
-                backward_k_min: int = 1, timeout_after_k_min: float = 1.0):
+                backward_k_min: int = 1, timeout_after_k_min: float = 1.0,
+                backward_stats: Optional[dict] = None):
         N, d = x.shape
         k_max = sel_logits.shape[1]
         x_cpu = x.detach().cpu()
@@ -293,6 +298,7 @@ class _RemoteMoECallBatched(torch.autograd.Function):
         ctx.backward_timeout = backward_timeout
         ctx.backward_k_min = backward_k_min
         ctx.timeout_after_k_min = timeout_after_k_min
+        ctx.backward_stats = backward_stats
         ctx.device = x.device
         return y.to(x.device)
 
@@ -341,13 +347,18 @@ class _RemoteMoECallBatched(torch.autograd.Function):
         finally:
             pool.shutdown(wait=False)
 
-        if ok_count[0] < min(ctx.backward_k_min, len(live_experts)):
+        missed = ok_count[0] < min(ctx.backward_k_min, len(live_experts))
+        if missed:
             # reference semantics: proceed anyway (failed experts contribute
             # zero grads), but surface that the backward quorum was missed
             logger.warning(
                 f"backward quorum missed: {ok_count[0]}/{len(live_experts)} experts "
                 f"returned grads (backward_k_min={ctx.backward_k_min})"
             )
+        if ctx.backward_stats is not None:
+            ctx.backward_stats["experts_ok"] += ok_count[0]
+            ctx.backward_stats["experts_dropped"] += len(live_experts) - ok_count[0]
+            ctx.backward_stats["quorum_missed"] += int(missed)
 
         return (dx.to(ctx.device, dtype=grad_y.dtype),
-                dlogits.to(ctx.device), None, None, None, None, None, None, None)
+                dlogits.to(ctx.device), None, None, None, None, None, None, None, None)

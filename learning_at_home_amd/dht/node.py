@@ -79,7 +79,7 @@ class DHTNode:
         )
         visited: set = set()
         in_flight: set = set()
-        best = {"value": None, "expiration": -float("inf")}
+        best = {"value": None, "expiration": -float("inf"), "holder": None}
         done_event = asyncio.Event()
 
         def next_target() -> Optional[int]:
@@ -125,6 +125,7 @@ class DHTNode:
                         value, expiration, nearest = reply
                         if value is not None and expiration is not None and expiration > best["expiration"]:
                             best["value"], best["expiration"] = value, expiration
+                            best["holder"] = target
                     else:
                         nearest = reply
                     for found_id, ep in nearest:
@@ -141,12 +142,13 @@ class DHTNode:
             w.cancel()
         live = {n: ep for n, ep in candidates.items() if not self.protocol.is_blacklisted(ep)}
         nearest = heapq.nsmallest(k, live, key=lambda n: n ^ key_int)
-        return [(nid, live[nid]) for nid in nearest], best["value"], best["expiration"]
+        return ([(nid, live[nid]) for nid in nearest], best["value"], best["expiration"],
+                best["holder"])
 
     async def find_nearest_nodes(
         self, key_id: int, k: Optional[int] = None
     ) -> List[Tuple[int, Tuple[str, int]]]:
-        nearest, _, _ = await self._traverse(int(key_id), k or self.k, find_value=False)
+        nearest, _, _, _ = await self._traverse(int(key_id), k or self.k, find_value=False)
         return nearest
 
     # ------------------------------------------------------------------
@@ -229,11 +231,20 @@ class DHTNode:
             if value is not None and expiration >= sufficient_expiration:
                 return value, expiration
 
-        _, best_value, best_expiration = await self._traverse(
+        nearest, best_value, best_expiration, holder = await self._traverse(
             int(key_id), self.k, find_value=True, sufficient_expiration=sufficient_expiration
         )
         if best_value is not None and best_expiration >= get_dht_time():
             self.protocol.cache.store(key_id, best_value, best_expiration)
+            # Kademlia read-repair (§2.3 caching): replicate the fetched value
+            # at the closest observed node that did NOT return it, so a key's
+            # replica set heals toward the ideal one after churn instead of
+            # relying solely on the owner's re-announce
+            for nid, ep in nearest:
+                if nid != holder and nid != int(self.node_id):
+                    asyncio.ensure_future(
+                        self.protocol.call_store(ep, key_id, best_value, best_expiration))
+                    break
             return best_value, best_expiration
         return None, None
 

@@ -164,3 +164,45 @@ class TestBlacklistRecovery:
                 reborn.shutdown()
         finally:
             a.shutdown()
+
+
+def test_read_repair_replicates_value_on_get():
+    """Kademlia read-repair: after a successful lookup,
+    the value is cached at the closest observed node that did NOT hold it —
+    a key's replica set heals after churn instead of waiting for the owner's
+    next re-announce."""
+    import asyncio
+
+    from learning_at_home_amd.dht.node import DHTNode
+    from learning_at_home_amd.dht.routing import DHTID
+    from learning_at_home_amd.utils import get_dht_time
+
+    async def main():
+        seed = await DHTNode.create()
+        nodes = [seed]
+        for _ in range(5):
+            nodes.append(await DHTNode.create(initial_peers=[("127.0.0.1", seed.port)]))
+        await asyncio.sleep(0.2)
+        key = "repair_me"
+        key_id = DHTID.generate(key)
+        exp = get_dht_time() + 60
+        # plant the value on exactly ONE node's storage (no normal replication)
+        holder = nodes[2]
+        holder.protocol.storage.store(key_id, "v", exp)
+
+        def replicas():
+            return sum(1 for n in nodes if n.protocol.storage.get(key_id)[0] is not None)
+
+        assert replicas() == 1
+        value, _ = await nodes[5].get(key)
+        assert value == "v"
+        # the fire-and-forget repair store needs a beat to land
+        for _ in range(30):
+            await asyncio.sleep(0.1)
+            if replicas() >= 2:
+                break
+        assert replicas() >= 2, "read-repair did not replicate the value"
+        for n in nodes:
+            n.shutdown()
+
+    asyncio.run(main())

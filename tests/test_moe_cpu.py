@@ -284,6 +284,12 @@ class TestFaultTolerance:
             bwd_s = time.monotonic() - t0
             assert bwd_s < 10.0, f"backward took {bwd_s:.1f}s — early exit did not engage"
             assert x.grad is not None and torch.isfinite(x.grad).all()
+            # dropped-grad accounting: the hung expert
+            # was dropped in the FORWARD (straggler), so the backward only
+            # revisits the 3 survivors — all of which returned grads
+            stats = moe.backward_stats
+            assert stats["experts_ok"] == 3, stats
+            assert stats["experts_dropped"] == 0 and stats["quorum_missed"] == 0
         finally:
             stop.set()
             t.join(2)
