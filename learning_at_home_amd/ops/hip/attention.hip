@@ -67,7 +67,7 @@ LAH_DEVINL void attn_block_id(int nqb, int BH, int& qb, long& bh) {
 // ---------------------------------------------------------------------------
 // forward: one wave = 32 q rows; block = 4 waves = 128 q rows of one (b,h)
 // ---------------------------------------------------------------------------
-template <int MINW, bool PREFETCH>
+template <int MINW, bool PREFETCH, int FKV>
 __launch_bounds__(256, MINW) __global__
 void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
                            const __bf16* __restrict__ K,   // [BH, S, 64]
@@ -75,7 +75,9 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
                            __bf16* __restrict__ O,
                            float* __restrict__ LSE,  // [BH, S] or null
                            int BH, int S, float scale2) {
-  __shared__ __attribute__((aligned(16))) __bf16 smem[4 * 32 * AT_LD];
+  constexpr int KV = FKV * 16;   // key-tile width (64 or 128)
+  constexpr int PS_LD = KV + 8;  // wave-private P row stride
+  __shared__ __attribute__((aligned(16))) __bf16 smem[4 * 32 * (KV + 8)];
 
   int qb;   // 128-row query block
   long bh;
@@ -85,7 +87,7 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
   const long base = bh * (long)S * 64;
   const long baseT = bh * (long)64 * S;  // Vt
   const int qw = qb * 128 + wave * 32;   // this wave's first q row
-  __bf16* Ps = smem + wave * 32 * AT_LD;
+  __bf16* Ps = smem + wave * 32 * PS_LD;
 
   // Q fragments, resident for the whole kernel: aq[m][kk]
   bf16x8 aq[2][2];
@@ -112,31 +114,31 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
 #pragma unroll
   for (int j = 0; j < 8; ++j) bones[j] = (l15 == 0) ? (__bf16)1.0f : (__bf16)0.0f;
 
-  const int kb_last = (qw + 31) >> 6;  // this wave's causal bound
+  const int kb_last = (qw + 31) / KV;  // this wave's causal bound
 
   // K fragments double-buffered in registers: tile kb+1's loads issue
   // BEFORE tile kb's softmax/PV, so their HBM/L2 latency hides under
   // ~1.5k cycles of compute (guide T14 async-STAGE, register form)
-  bf16x8 bkA[2][4], bkB[2][4];
-  auto load_k = [&](bf16x8(&dst)[2][4], int kb_) {
+  bf16x8 bkA[2][FKV], bkB[2][FKV];
+  auto load_k = [&](bf16x8(&dst)[2][FKV], int kb_) {
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-      for (int f = 0; f < 4; ++f)
-        dst[kk][f] = *(const bf16x8*)(K + base + (long)(kb_ * 64 + f * 16 + l15) * 64 + kk * 32 + lk8);
+      for (int f = 0; f < FKV; ++f)
+        dst[kk][f] = *(const bf16x8*)(K + base + (long)(kb_ * KV + f * 16 + l15) * 64 + kk * 32 + lk8);
   };
 
-  auto tile = [&](const bf16x8(&bk)[2][4], int kb) {
+  auto tile = [&](const bf16x8(&bk)[2][FKV], int kb) {
     // ---- S = Q @ K^T ----------------------------------------------------
-    f32x4 sf[2][4];
+    f32x4 sf[2][FKV];
 #pragma unroll
     for (int m = 0; m < 2; ++m)
 #pragma unroll
-      for (int f = 0; f < 4; ++f) sf[m][f] = {0.f, 0.f, 0.f, 0.f};
+      for (int f = 0; f < FKV; ++f) sf[m][f] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-      for (int f = 0; f < 4; ++f) {
+      for (int f = 0; f < FKV; ++f) {
         sf[0][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[0][kk], bk[kk][f], sf[0][f], 0, 0, 0);
         sf[1][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(aq[1][kk], bk[kk][f], sf[1][f], 0, 0, 0);
       }
@@ -146,16 +148,16 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
     for (int m = 0; m < 2; ++m)
 #pragma unroll
       for (int r = 0; r < 4; ++r) rmax[m][r] = NEG_BIG;
-    const bool boundary = (kb * 64 + 63) > qw;  // wave-uniform
+    const bool boundary = (kb * KV + KV - 1) > qw;  // wave-uniform
     if (boundary) {
 #pragma unroll
       for (int m = 0; m < 2; ++m)
 #pragma unroll
-        for (int f = 0; f < 4; ++f)
+        for (int f = 0; f < FKV; ++f)
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             int qg = qw + m * 16 + r0 + r;
-            int kg = kb * 64 + f * 16 + l15;
+            int kg = kb * KV + f * 16 + l15;
             sf[m][f][r] = (kg <= qg) ? sf[m][f][r] * scale2 : NEG_BIG;
             rmax[m][r] = fmaxf(rmax[m][r], sf[m][f][r]);
           }
@@ -163,7 +165,7 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
 #pragma unroll
       for (int m = 0; m < 2; ++m)
 #pragma unroll
-        for (int f = 0; f < 4; ++f)
+        for (int f = 0; f < FKV; ++f)
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             sf[m][f][r] *= scale2;
@@ -201,19 +203,19 @@ void flash_attn_fwd_kernel(const __bf16* __restrict__ Q,   // [BH, S, 64]
 #pragma unroll
     for (int m = 0; m < 2; ++m)
 #pragma unroll
-      for (int f = 0; f < 4; ++f)
+      for (int f = 0; f < FKV; ++f)
 #pragma unroll
         for (int r = 0; r < 4; ++r)
-          Ps[(m * 16 + r0 + r) * AT_LD + f * 16 + l15] =
+          Ps[(m * 16 + r0 + r) * PS_LD + f * 16 + l15] =
               f2bf(fast_exp2(sf[m][f][r] - m_run[m][r]));
     // ---- [O | l] += P @ [V | 1]: A = P rows (LDS), B = Vt rows (global) --
 #pragma unroll
-    for (int kk = 0; kk < 2; ++kk) {
-      bf16x8 ap0 = *(const bf16x8*)(Ps + (0 * 16 + l15) * AT_LD + kk * 32 + lk8);
-      bf16x8 ap1 = *(const bf16x8*)(Ps + (1 * 16 + l15) * AT_LD + kk * 32 + lk8);
+    for (int kk = 0; kk < KV / 32; ++kk) {
+      bf16x8 ap0 = *(const bf16x8*)(Ps + (0 * 16 + l15) * PS_LD + kk * 32 + lk8);
+      bf16x8 ap1 = *(const bf16x8*)(Ps + (1 * 16 + l15) * PS_LD + kk * 32 + lk8);
 #pragma unroll
       for (int f = 0; f < 4; ++f) {
-        bf16x8 bv = *(const bf16x8*)(Vt + baseT + (long)(f * 16 + l15) * S + kb * 64 + kk * 32 + lk8);
+        bf16x8 bv = *(const bf16x8*)(Vt + baseT + (long)(f * 16 + l15) * S + kb * KV + kk * 32 + lk8);
         of[0][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap0, bv, of[0][f], 0, 0, 0);
         of[1][f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ap1, bv, of[1][f], 0, 0, 0);
       }
@@ -496,9 +498,10 @@ extern "C" void lah_flash_attn_fwd(const void* Q, const void* K, const void* V,
     const char* e = getenv("LAH_ATTN_V");
     v = e ? atoi(e) : 0;
   }
-  auto kern = flash_attn_fwd_kernel<2, true>;       // v0: 2 waves + K prefetch
-  if (v == 1) kern = flash_attn_fwd_kernel<3, false>;  // v1: 3 waves, no prefetch
-  if (v == 2) kern = flash_attn_fwd_kernel<3, true>;   // v2: 3 waves + prefetch (spills)
+  auto kern = flash_attn_fwd_kernel<2, true, 4>;       // v0: 2 waves + K prefetch
+  if (v == 1) kern = flash_attn_fwd_kernel<3, false, 4>;  // v1: 3 waves, no prefetch
+  if (v == 2) kern = flash_attn_fwd_kernel<3, true, 4>;   // v2: 3 waves + prefetch (spills)
+  if (v == 3) kern = flash_attn_fwd_kernel<2, false, 8>;  // v3: 128-wide KV tiles
   hipLaunchKernelGGL(kern, grid, block, 0, stream,
                      (const __bf16*)Q, (const __bf16*)K, (const __bf16*)Vt_scratch,
                      (__bf16*)O, (float*)LSE, BH, S, scale2);
