@@ -7,8 +7,11 @@ logits live only transiently; the backward recomputes the chunk's softmax
 and feeds dX / dW GEMMs directly. Library GEMMs (hipBLASLt) — this is a
 torch-level fusion, not a kernel.
 
-Numerics: per-chunk logits in fp32 via bf16 GEMM (same as the unfused path);
-loss is the exact valid-token-mean CE. Labels equal to `ignore_index`
+Numerics: per-chunk logits from the bf16 GEMM, read by the CE kernels in
+their native dtype with fp32 lane conversion + fp32 accumulation — bitwise
+the same result as materializing an fp32 copy first (the values are already
+bf16-rounded), without the cast kernel or the fp32 [chunk, V] traffic.
+Loss is the exact valid-token-mean CE. Labels equal to `ignore_index`
 (default -100, matching F.cross_entropy) contribute neither to the loss
 nor to the gradients, and the mean is over VALID tokens only — identical
 semantics to the unfused `F.cross_entropy(..., ignore_index=-100)` path.
@@ -21,7 +24,8 @@ from . import _C, HAS_HIP
 
 
 def _use_ce_kernels(x: torch.Tensor, weight: torch.Tensor) -> bool:
-    return HAS_HIP and x.is_cuda and weight.shape[0] % 4 == 0
+    return (HAS_HIP and x.is_cuda and weight.shape[0] % 4 == 0
+            and x.dtype in (torch.bfloat16, torch.float32))
 
 
 class _ChunkedLinearCE(torch.autograd.Function):
@@ -36,18 +40,22 @@ class _ChunkedLinearCE(torch.autograd.Function):
         lse_all = torch.empty(T, dtype=torch.float32, device=x.device) if use_hip else None
         for s in range(0, T, chunk):
             xs = x[s : s + chunk]
-            logits = (xs @ weight.t()).float()
             if use_hip:
-                # one logits read: online lse + label gather (ce.hip); the
-                # torch path materializes log-softmax over [chunk, V] instead
+                # one logits read: online lse + label gather (ce.hip) straight
+                # off the GEMM's native dtype (bf16 weights → bf16 logits; the
+                # kernel converts lanes to fp32 — identical numerics to an
+                # explicit .float() materialization, minus the cast kernel and
+                # the fp32 [chunk, V] traffic); the torch path materializes
+                # log-softmax over [chunk, V] fp32 instead
+                logits = xs @ weight.t()
                 lse, picked = _C.ce_rowstats(logits, labels_i32[s : s + chunk],
                                              ignore_index)
                 lse_all[s : s + chunk] = lse
                 total += (lse - picked).sum()
             else:
                 total += torch.nn.functional.cross_entropy(
-                    logits, labels[s : s + chunk], reduction="sum",
-                    ignore_index=ignore_index,
+                    (xs @ weight.t()).float(), labels[s : s + chunk],
+                    reduction="sum", ignore_index=ignore_index,
                 )
         ctx.save_for_backward(x, weight, labels,
                               lse_all if use_hip else torch.empty(0))
@@ -72,7 +80,7 @@ class _ChunkedLinearCE(torch.autograd.Function):
             gs = float(dloss) / max(ctx.n_valid, 1)
             for s in range(0, T, chunk):
                 xs = x[s : s + chunk]
-                logits = (xs @ weight.t()).float()
+                logits = xs @ weight.t()  # native dtype; kernel reads bf16/fp32
                 dlogits = _C.ce_bwd(logits, labels_i32[s : s + chunk],
                                     lse_all[s : s + chunk].contiguous(), gs,
                                     ignore_index).to(x.dtype)

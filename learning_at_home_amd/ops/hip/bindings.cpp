@@ -36,9 +36,10 @@ void lah_sgd_step_t(void*, const void*, void*, void*, int, int, int, float, floa
 void lah_sgd_step_sr_t(void*, const void*, void*, int, int, int, float, float,
                        unsigned, hipStream_t);
 void lah_transpose_bf16(const void*, void*, int, int, int, hipStream_t);
-void lah_ce_rowstats(const void*, const void*, void*, void*, int, long, int, hipStream_t);
+void lah_ce_rowstats(const void*, const void*, void*, void*, int, long, int, int,
+                     hipStream_t);
 void lah_ce_bwd(const void*, const void*, const void*, void*, float, int, long, int,
-                hipStream_t);
+                int, hipStream_t);
 void lah_colsum(const void*, void*, int, int, int, hipStream_t);
 }
 
@@ -115,29 +116,34 @@ void fused_dw_sgd_sr(torch::Tensor A, torch::Tensor B, torch::Tensor W,
 // chunked lm_head CE row stats: (lse, picked) per row, one logits read
 std::vector<torch::Tensor> ce_rowstats(torch::Tensor logits, torch::Tensor labels,
                                        int64_t ignore_index) {
-  CHECK_CUDA(logits); CHECK_F32(logits); CHECK_CONTIG(logits);
+  CHECK_CUDA(logits); CHECK_CONTIG(logits);
+  bool bf16 = logits.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(bf16 || logits.scalar_type() == at::kFloat, "logits must be bf16 or fp32");
   CHECK_I32(labels); CHECK_CONTIG(labels);
   int T = logits.size(0);
   long V = logits.size(1);
   TORCH_CHECK(labels.size(0) == T, "labels/logits mismatch");
   TORCH_CHECK(V % 4 == 0, "V must be a multiple of 4");
-  auto lse = torch::empty({T}, logits.options());
-  auto picked = torch::empty({T}, logits.options());
+  auto lse = torch::empty({T}, logits.options().dtype(at::kFloat));
+  auto picked = torch::empty({T}, logits.options().dtype(at::kFloat));
   lah_ce_rowstats(logits.data_ptr(), labels.data_ptr(), lse.data_ptr(),
-                  picked.data_ptr(), T, V, (int)ignore_index, cur_stream());
+                  picked.data_ptr(), T, V, (int)ignore_index, (int)bf16,
+                  cur_stream());
   return {lse, picked};
 }
 
 torch::Tensor ce_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
                      double gscale, int64_t ignore_index) {
-  CHECK_CUDA(logits); CHECK_F32(logits); CHECK_CONTIG(logits);
+  CHECK_CUDA(logits); CHECK_CONTIG(logits);
+  bool bf16 = logits.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(bf16 || logits.scalar_type() == at::kFloat, "logits must be bf16 or fp32");
   CHECK_I32(labels); CHECK_CONTIG(labels); CHECK_F32(lse); CHECK_CONTIG(lse);
   int T = logits.size(0);
   long V = logits.size(1);
   TORCH_CHECK(V % 4 == 0, "V must be a multiple of 4");
   auto d = torch::empty({T, V}, logits.options().dtype(at::kBFloat16));
   lah_ce_bwd(logits.data_ptr(), labels.data_ptr(), lse.data_ptr(), d.data_ptr(),
-             (float)gscale, T, V, (int)ignore_index, cur_stream());
+             (float)gscale, T, V, (int)ignore_index, (int)bf16, cur_stream());
   return d;
 }
 

@@ -8,27 +8,34 @@
 //   bwd:  dlogits = (softmax - onehot) * gscale, bf16, zero for ignored rows
 // The lse vector ([T] fp32) is saved by the autograd ctx, so the backward
 // needs no max/sum recompute — one GEMM + one elementwise pass per chunk.
+//
+// Templated on the logits dtype: the training path feeds the lm_head GEMM's
+// bf16 output DIRECTLY (reading bf16 here is numerics-identical to casting
+// to fp32 first — the values are already bf16-rounded — and skips both the
+// 2×[T,V] fp32 materialization and its cast kernel, ~11 ms/step at the
+// flagship shape); fp32 logits remain supported for fp32 callers.
 #include "common.h"
 
 #define CE_LOG2E 1.4426950408889634f
 #define CE_LN2 0.6931471805599453f
 
+template <typename TS, typename T4>
 __launch_bounds__(256) __global__
-void ce_rowstats_kernel(const float* __restrict__ logits, const int* __restrict__ labels,
+void ce_rowstats_kernel(const TS* __restrict__ logits, const int* __restrict__ labels,
                         float* __restrict__ lse, float* __restrict__ picked,
                         int ignore_index, long V) {
   const int row = blockIdx.x;
-  const float* lr = logits + (long)row * V;
+  const TS* lr = logits + (long)row * V;
   const int tid = threadIdx.x;
   __shared__ float sm[8], ss[8];
 
-  // per-thread online max/sum in log2 domain over a strided f32x4 sweep
+  // per-thread online max/sum in log2 domain over a strided 4-wide sweep
   float m = -3.0e38f, s = 0.f;
   for (long c = (long)tid * 4; c + 3 < V; c += 256 * 4) {
-    f32x4 v = *(const f32x4*)(lr + c);
+    T4 v = *(const T4*)(lr + c);
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      float l2 = v[j] * CE_LOG2E;
+      float l2 = (float)v[j] * CE_LOG2E;
       if (l2 > m) {
         s = s * __builtin_amdgcn_exp2f(m - l2) + 1.0f;
         m = l2;
@@ -60,12 +67,13 @@ void ce_rowstats_kernel(const float* __restrict__ logits, const int* __restrict_
     float l = (mt + __builtin_amdgcn_logf(st)) * CE_LN2;  // natural log
     lse[row] = l;
     int lab = labels[row];
-    picked[row] = (lab == ignore_index || lab < 0 || lab >= V) ? l : lr[lab];
+    picked[row] = (lab == ignore_index || lab < 0 || lab >= V) ? l : (float)lr[lab];
   }
 }
 
+template <typename TS, typename T4>
 __launch_bounds__(256) __global__
-void ce_bwd_kernel(const float* __restrict__ logits, const int* __restrict__ labels,
+void ce_bwd_kernel(const TS* __restrict__ logits, const int* __restrict__ labels,
                    const float* __restrict__ lse, __bf16* __restrict__ dlogits,
                    float gscale, int ignore_index, long V) {
   const int row = blockIdx.y;
@@ -74,15 +82,15 @@ void ce_bwd_kernel(const float* __restrict__ logits, const int* __restrict__ lab
   const int lab = labels[row];
   const bool ignored = (lab == ignore_index || lab < 0);
   const float l2 = lse[row] * CE_LOG2E;
-  const float* lr = logits + (long)row * V;
+  const TS* lr = logits + (long)row * V;
   __bf16* dr = dlogits + (long)row * V;
   if (c0 + 3 < V) {
-    f32x4 v = *(const f32x4*)(lr + c0);
+    T4 v = *(const T4*)(lr + c0);
     bf16x4 o;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       float p = ignored ? 0.f
-                        : __builtin_amdgcn_exp2f(v[j] * CE_LOG2E - l2) * gscale
+                        : __builtin_amdgcn_exp2f((float)v[j] * CE_LOG2E - l2) * gscale
                               - ((c0 + j == lab) ? gscale : 0.f);
       o[j] = f2bf(p);
     }
@@ -90,7 +98,7 @@ void ce_bwd_kernel(const float* __restrict__ logits, const int* __restrict__ lab
   } else {
     for (long c = c0; c < V; ++c) {
       float p = ignored ? 0.f
-                        : __builtin_amdgcn_exp2f(lr[c] * CE_LOG2E - l2) * gscale
+                        : __builtin_amdgcn_exp2f((float)lr[c] * CE_LOG2E - l2) * gscale
                               - ((c == lab) ? gscale : 0.f);
       dr[c] = f2bf(p);
     }
@@ -99,17 +107,27 @@ void ce_bwd_kernel(const float* __restrict__ logits, const int* __restrict__ lab
 
 extern "C" void lah_ce_rowstats(const void* logits, const void* labels, void* lse,
                                 void* picked, int T, long V, int ignore_index,
-                                hipStream_t stream) {
-  hipLaunchKernelGGL(ce_rowstats_kernel, dim3(T), dim3(256), 0, stream,
-                     (const float*)logits, (const int*)labels, (float*)lse,
-                     (float*)picked, ignore_index, V);
+                                int logits_bf16, hipStream_t stream) {
+  if (logits_bf16)
+    hipLaunchKernelGGL((ce_rowstats_kernel<__bf16, bf16x4>), dim3(T), dim3(256), 0,
+                       stream, (const __bf16*)logits, (const int*)labels,
+                       (float*)lse, (float*)picked, ignore_index, V);
+  else
+    hipLaunchKernelGGL((ce_rowstats_kernel<float, f32x4>), dim3(T), dim3(256), 0,
+                       stream, (const float*)logits, (const int*)labels,
+                       (float*)lse, (float*)picked, ignore_index, V);
 }
 
 extern "C" void lah_ce_bwd(const void* logits, const void* labels, const void* lse,
                            void* dlogits, float gscale, int T, long V,
-                           int ignore_index, hipStream_t stream) {
+                           int ignore_index, int logits_bf16, hipStream_t stream) {
   dim3 grid((unsigned)((V + 1023) / 1024), T), block(256);
-  hipLaunchKernelGGL(ce_bwd_kernel, grid, block, 0, stream, (const float*)logits,
-                     (const int*)labels, (const float*)lse, (__bf16*)dlogits,
-                     gscale, ignore_index, V);
+  if (logits_bf16)
+    hipLaunchKernelGGL((ce_bwd_kernel<__bf16, bf16x4>), grid, block, 0, stream,
+                       (const __bf16*)logits, (const int*)labels, (const float*)lse,
+                       (__bf16*)dlogits, gscale, ignore_index, V);
+  else
+    hipLaunchKernelGGL((ce_bwd_kernel<float, f32x4>), grid, block, 0, stream,
+                       (const float*)logits, (const int*)labels, (const float*)lse,
+                       (__bf16*)dlogits, gscale, ignore_index, V);
 }

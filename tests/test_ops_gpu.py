@@ -576,6 +576,47 @@ class TestFusedCEKernels:
         ref = probs * gs
         assert_close_bf16(d, ref, rtol=1e-2, atol=1e-4, what="ce_bwd")
 
+    def test_rowstats_bf16_bitwise_matches_fp32_cast(self):
+        """Reading bf16 logits directly must equal casting them to fp32 first
+        (the cast is exact), for lse, picked and the backward dlogits."""
+        torch.manual_seed(43)
+        T, V = 222, 1024
+        logits16 = (torch.randn(T, V, device=DEV) * 3).bfloat16().contiguous()
+        logits32 = logits16.float()
+        labels = torch.randint(0, V, (T,), device=DEV).int()
+        labels[::9] = -100
+        lse16, picked16 = _C.ce_rowstats(logits16, labels, -100)
+        lse32, picked32 = _C.ce_rowstats(logits32, labels, -100)
+        assert torch.equal(lse16, lse32)
+        assert torch.equal(picked16, picked32)
+        gs = 1.0 / 197
+        d16 = _C.ce_bwd(logits16, labels, lse16.contiguous(), gs, -100)
+        d32 = _C.ce_bwd(logits32, labels, lse32.contiguous(), gs, -100)
+        assert torch.equal(d16, d32)
+
+    def test_chunked_ce_bf16_end_to_end(self):
+        """bf16 x/w through the HIP chunked CE vs unfused F.cross_entropy on
+        the same bf16 logits cast to fp32 (loss + grads)."""
+        from learning_at_home_amd.ops.fused_ce import chunked_linear_cross_entropy
+
+        torch.manual_seed(44)
+        T, D, V = 192, 64, 512
+        x = torch.randn(T, D, device=DEV).bfloat16().requires_grad_(True)
+        w = torch.randn(V, D, device=DEV).bfloat16().requires_grad_(True)
+        labels = torch.randint(0, V, (T,), device=DEV)
+        labels[::6] = -100
+        loss = chunked_linear_cross_entropy(x, w, labels, chunk=64)
+        loss.backward()
+
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        ref = torch.nn.functional.cross_entropy(
+            (x2 @ w2.t()).float(), labels, ignore_index=-100)
+        ref.backward()
+        assert torch.allclose(loss, ref.float(), rtol=1e-4, atol=1e-4)
+        assert torch.allclose(x.grad.float(), x2.grad.float(), rtol=5e-2, atol=2e-3)
+        assert torch.allclose(w.grad.float(), w2.grad.float(), rtol=5e-2, atol=2e-3)
+
     def test_chunked_ce_gpu_matches_cpu_path(self):
         """End-to-end chunked CE: HIP kernel path vs the torch fallback on
         identical fp32 inputs (loss and both grads)."""
