@@ -149,17 +149,17 @@ class TestExpertParallelGloo:
         mp.spawn(_ep_worker, args=(2, str(tmp_path)), nprocs=2, join=True)
 
 
-def _chunked_worker(rank, world):
+def _chunked_worker(rank, world, num_experts=8, n_chunks=2):
     os.environ.setdefault("MASTER_ADDR", "127.0.0.1")  # port set by parent
     dist.init_process_group("gloo", rank=rank, world_size=world)
     torch.manual_seed(7)
-    plain = ExpertParallelMoE(d_model=16, ffn_dim=32, num_experts=8, k=2,
+    plain = ExpertParallelMoE(d_model=16, ffn_dim=32, num_experts=num_experts, k=2,
                               capacity_factor=4.0, aux_loss_coef=0.01,
                               a2a_chunks=1)
     torch.manual_seed(7)
-    chunked = ExpertParallelMoE(d_model=16, ffn_dim=32, num_experts=8, k=2,
+    chunked = ExpertParallelMoE(d_model=16, ffn_dim=32, num_experts=num_experts, k=2,
                                 capacity_factor=4.0, aux_loss_coef=0.01,
-                                a2a_chunks=2)
+                                a2a_chunks=n_chunks)
     plain.eval(); chunked.eval()
     torch.manual_seed(200 + rank)
     x = torch.randn(10, 16, requires_grad=True)
@@ -204,6 +204,13 @@ class TestChunkedA2AOverlap:
     def test_two_rank_gloo_chunked_matches_unchunked(self):
         os.environ["MASTER_PORT"] = str(_free_port())
         mp.spawn(_chunked_worker, args=(2,), nprocs=2, join=True)
+
+    def test_four_rank_gloo_chunked_uneven_ranges(self):
+        """4 gloo ranks, P=3 local experts, 2 chunks → uneven ranges (2+1):
+        exercises the W>2 dispatch/return reshapes and non-divisible chunk
+        bounds the driver's 8-GPU run will hit (e.g. E=256, N=8, chunks=4)."""
+        os.environ["MASTER_PORT"] = str(_free_port())
+        mp.spawn(_chunked_worker, args=(4, 12, 2), nprocs=4, join=True)
 
     def test_slice_grad_ready_ranges(self):
         """The stale-accumulate pitfall: after zero_main_grads, each chunk's
