@@ -117,3 +117,112 @@ class TestTimedStorageProperties:
         for key, (val, exp) in best.items():
             got_val, got_exp = store.get(key)
             assert got_exp == exp
+
+
+class TestRoutingOpsProperties:
+    """Invariants of the CPU routing-op fallbacks (identical semantics to the
+    HIP kernels — the GPU suite checks kernel-vs-fallback equality, these
+    check the fallback against the mathematical contract)."""
+
+    @given(st.integers(1, 24), st.integers(2, 12), st.sampled_from([1, 2, 4]),
+           st.integers(0, 2**31 - 1))
+    @settings(max_examples=40, deadline=None)
+    def test_gating_topk_simplex_and_uniqueness(self, T, E, k, seed):
+        from learning_at_home_amd.ops.functional import gating_topk
+        if k > E:
+            return
+        torch.manual_seed(seed)
+        logits = torch.randn(T, E)
+        idx, w, probs = gating_topk(logits, k)
+        assert idx.shape == (T, k) and w.shape == (T, k)
+        for t in range(T):
+            assert len(set(idx[t].tolist())) == k  # k distinct experts
+        assert torch.all(w >= 0)
+        assert torch.allclose(w.sum(-1), torch.ones(T), atol=1e-5)
+        assert torch.allclose(probs.float().sum(-1), torch.ones(T), atol=1e-4)
+        # selected experts are exactly the top-k by logit value
+        ref = logits.topk(k, dim=-1).indices.sort(-1).values
+        assert torch.equal(idx.long().sort(-1).values, ref)
+
+    @given(st.integers(1, 30), st.sampled_from([1, 2, 4]), st.integers(2, 8),
+           st.integers(1, 6), st.integers(0, 2**31 - 1))
+    @settings(max_examples=40, deadline=None)
+    def test_assign_slots_invariants(self, T, k, E, C, seed):
+        from learning_at_home_amd.ops.functional import assign_slots
+        torch.manual_seed(seed)
+        topk_idx = torch.randint(0, E, (T, k), dtype=torch.int32)
+        pos, counts, inv = assign_slots(topk_idx, E, C)
+        assert torch.equal(counts.long(),
+                           torch.bincount(topk_idx.reshape(-1).long(), minlength=E))
+        # every kept (t, j) occupies a unique in-bounds slot of ITS expert,
+        # and inv is its exact inverse
+        seen = set()
+        for t in range(T):
+            for j in range(k):
+                p = int(pos[t, j])
+                e = int(topk_idx[t, j])
+                if p < 0:
+                    assert int(counts[e]) > C  # only over-capacity drops
+                    continue
+                assert 0 <= p < C
+                assert (e, p) not in seen
+                seen.add((e, p))
+                assert int(inv[e, p]) == t * k + j
+        # every non-(-1) inv entry round-trips
+        for e in range(E):
+            kept = int(min(int(counts[e]), C))
+            assert (inv[e, :kept] >= 0).all()
+            assert (inv[e, kept:] == -1).all()
+
+    @given(st.integers(1, 12), st.integers(2, 6), st.sampled_from([1, 2]),
+           st.integers(0, 2**31 - 1))
+    @settings(max_examples=25, deadline=None)
+    def test_dispatch_combine_roundtrip_no_drops(self, T, E, k, seed):
+        """With capacity >= T*k nothing drops, so combine(dispatch(x)) must be
+        exactly the gate-weighted sum of (k copies of) each token."""
+        from learning_at_home_amd.ops.functional import (assign_slots, combine,
+                                                         dispatch, gating_topk)
+        if k > E:
+            return
+        torch.manual_seed(seed)
+        D, C = 8, T * k
+        x = torch.randn(T, D)
+        logits = torch.randn(T, E)
+        idx, w, _ = gating_topk(logits, k)
+        pos, counts, inv = assign_slots(idx, E, C)
+        assert (pos >= 0).all()
+        xbuf = dispatch(x, idx, pos, inv, k)
+        y = combine(xbuf, idx, pos, w, inv)
+        ref = (w.sum(-1, keepdim=True)) * x  # sum_j w_j * x (same token back)
+        assert torch.allclose(y, ref, atol=1e-5), (y - ref).abs().max().item()
+
+
+class TestChunkedCEProperties:
+    @given(st.integers(1, 40), st.sampled_from([4, 8, 12]), st.integers(1, 45),
+           st.floats(0.0, 1.0), st.integers(0, 2**31 - 1))
+    @settings(max_examples=40, deadline=None)
+    def test_matches_unfused_any_chunk_and_ignores(self, T, V, chunk, ign_frac, seed):
+        """Loss and grads equal F.cross_entropy for ANY chunk size and ANY
+        ignore pattern (including all tokens ignored)."""
+        import torch.nn.functional as F
+        from learning_at_home_amd.ops.fused_ce import chunked_linear_cross_entropy
+        torch.manual_seed(seed)
+        D = 6
+        x = torch.randn(T, D, requires_grad=True)
+        w = torch.randn(V, D, requires_grad=True)
+        labels = torch.randint(0, V, (T,))
+        labels[torch.rand(T) < ign_frac] = -100
+        loss = chunked_linear_cross_entropy(x, w, labels, chunk=chunk)
+        loss.backward()
+        assert torch.isfinite(loss)
+        x2 = x.detach().clone().requires_grad_(True)
+        w2 = w.detach().clone().requires_grad_(True)
+        if (labels != -100).any():
+            ref = F.cross_entropy((x2 @ w2.t()).float(), labels, ignore_index=-100)
+            ref.backward()
+            assert torch.allclose(loss, ref, rtol=1e-5, atol=1e-5)
+            assert torch.allclose(x.grad, x2.grad, rtol=1e-4, atol=1e-5)
+            assert torch.allclose(w.grad, w2.grad, rtol=1e-4, atol=1e-5)
+        else:
+            assert loss.item() == 0.0
+            assert torch.all(x.grad == 0) and torch.all(w.grad == 0)
