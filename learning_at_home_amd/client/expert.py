@@ -13,7 +13,7 @@ from __future__ import annotations
 import queue
 import socket
 import threading
-from typing import Dict, List, Optional, Sequence, Tuple
+from typing import Dict, Optional, Tuple
 
 import torch
 import torch.nn as nn

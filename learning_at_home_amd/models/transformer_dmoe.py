@@ -12,7 +12,7 @@ mapping hidden states [B, S, D] → [B, S, D], so the same trunk runs over
 from __future__ import annotations
 
 import os
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Callable, Optional
 
 import torch

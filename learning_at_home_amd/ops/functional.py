@@ -11,7 +11,6 @@ and the two weight-gradient TN GEMMs writing fp32 grads in place
 """
 from __future__ import annotations
 
-from typing import Tuple
 
 import torch
 import torch.nn.functional as F

@@ -5,7 +5,7 @@ CollaborativeOptimizer shape: wrap any torch optimizer; every
 parameters (asynchronously — training never blocks on a barrier)."""
 from __future__ import annotations
 
-from typing import Optional, Sequence, Union
+from typing import Optional, Union
 
 import torch
 import torch.distributed as dist

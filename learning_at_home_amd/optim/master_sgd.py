@@ -7,7 +7,7 @@ fp32, and re-quantizes the bf16 parameter — the trunk-side analog of the
 experts' fused master-SGD (parallel/FusedExpertSGD)."""
 from __future__ import annotations
 
-from typing import Iterable, Optional
+from typing import Iterable
 
 import torch
 

@@ -15,7 +15,7 @@ CDNA4 grouped-FFN kernels (ops/hip) — see parallel/expert_shard.py.
 """
 from __future__ import annotations
 
-from typing import Any, Callable, Dict, Optional, Sequence, Tuple
+from typing import Any, Dict, Optional, Sequence, Tuple
 
 import torch
 import torch.nn as nn

@@ -2,7 +2,7 @@
 `get_info` — clients fetch these to build correctly-shaped RPC payloads)."""
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Tuple
 
 import torch
